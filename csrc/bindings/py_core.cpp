@@ -341,6 +341,24 @@ PYBIND11_MODULE(_core, m) {
       .def_property_readonly("is_leader", &coord::LeaderElector::is_leader)
       .def("current_leader", &coord::LeaderElector::current_leader);
 
+  // Test-only: serialize a one-shard DATA_PULL request exactly as the wire
+  // sees it (hostile-input tests craft out-of-range pulls without mirroring
+  // the serde layout in Python).
+  m.def("encode_pull_req_for_test",
+        [](const std::string& dst_pool, uint64_t dst_offset, uint64_t total_len,
+           const std::string& src_pool, uint64_t src_offset, uint64_t src_len) {
+          ShardPlacement sp;
+          sp.pool_id = src_pool;
+          sp.offset = src_offset;
+          sp.length = src_len;
+          serde::Enc e;
+          serde::put(e, dst_pool);
+          serde::put(e, dst_offset);
+          serde::put(e, total_len);
+          serde::put(e, std::vector<ShardPlacement>{sp});
+          return py::bytes(e.buf);
+        });
+
   // ------------------------------------------------------------- gpu
   auto gm = m.def_submodule("gpu");
   gm.def("available", &gpu::available);

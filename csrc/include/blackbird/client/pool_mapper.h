@@ -26,11 +26,15 @@ class PoolMapper {
   }
 
   // SHM: returns mapped base or nullptr (maps the whole segment; size taken
-  // from the segment itself).
-  void* map_shm(const std::string& name, uint64_t size_hint) {
+  // from the segment itself, reported via *mapped_size when non-null).
+  void* map_shm(const std::string& name, uint64_t size_hint,
+                uint64_t* mapped_size = nullptr) {
     std::lock_guard<std::mutex> g(mu_);
     auto it = shm_.find(name);
-    if (it != shm_.end()) return it->second.ptr;
+    if (it != shm_.end()) {
+      if (mapped_size) *mapped_size = it->second.size;
+      return it->second.ptr;
+    }
     int fd = shm_open(name.c_str(), O_RDWR, 0600);
     if (fd < 0) return nullptr;
     struct stat st {};
@@ -45,6 +49,7 @@ class PoolMapper {
     ::close(fd);
     if (p == MAP_FAILED) return nullptr;
     shm_[name] = {p, size};
+    if (mapped_size) *mapped_size = size;
     return p;
   }
 

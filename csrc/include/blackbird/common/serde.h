@@ -37,6 +37,7 @@ class Dec {
  public:
   Dec(const void* p, size_t n) : p_(static_cast<const char*>(p)), end_(p_ + n) {}
   bool ok() const { return ok_; }
+  void fail() { ok_ = false; }
   size_t remaining() const { return static_cast<size_t>(end_ - p_); }
 
   bool raw(void* out, size_t n) {
@@ -103,8 +104,9 @@ template <typename T>
 void get(Dec& d, std::vector<T>& v) {
   uint32_t n = d.num<uint32_t>();
   v.clear();
-  // Guard against hostile/corrupt lengths: each element needs ≥1 byte.
-  if (n > d.remaining() && n > (1u << 24)) return;
+  // Guard against hostile/corrupt lengths: each element needs ≥1 byte. The
+  // rejection must surface as a decode error, not as a valid empty vector.
+  if (n > d.remaining() && n > (1u << 24)) { d.fail(); return; }
   v.reserve(n);
   for (uint32_t i = 0; i < n && d.ok(); ++i) {
     T x{};

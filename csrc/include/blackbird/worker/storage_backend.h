@@ -105,9 +105,9 @@ class LocalPools {
   void add(const PoolId& id, void* base, uint64_t size, bool is_device,
            int device);
   void remove(const PoolId& id);
-  // returns base or nullptr; *is_device set when found
+  // returns base or nullptr; *is_device/*device/*size set when found
   void* lookup(const PoolId& id, bool* is_device = nullptr,
-               int* device = nullptr);
+               int* device = nullptr, uint64_t* size = nullptr);
 
  private:
   struct Entry {

@@ -1368,13 +1368,19 @@ void KeystoneService::cleanup_dead_worker(const WorkerId& id) {
     for (auto& [key, meta] : objects_) {
       auto& copies = meta.copies;
       auto before = copies.size();
-      copies.erase(std::remove_if(copies.begin(), copies.end(),
-                                  [&](const CopyPlacement& c) {
-                                    for (const auto& s : c.shards)
-                                      if (s.worker_id == id) return true;
-                                    return false;
-                                  }),
-                   copies.end());
+      auto dead = std::stable_partition(copies.begin(), copies.end(),
+                                        [&](const CopyPlacement& c) {
+                                          for (const auto& s : c.shards)
+                                            if (s.worker_id == id) return false;
+                                          return true;
+                                        });
+      // Trim the matching leases from the allocator ledger too: if the worker
+      // re-registers the same pool_id with a fresh PoolAllocator, a later
+      // free/expiry must not replay stale leases into the new allocator
+      // (double allocation / freeing live ranges).
+      for (auto it = dead; it != copies.end(); ++it)
+        allocator_.free_ranges(key, it->shards);
+      copies.erase(dead, copies.end());
       if (copies.empty() && before > 0) lost.push_back(key);
     }
     for (const auto& k : lost) remove_object_locked(k);

@@ -124,12 +124,14 @@ void LocalPools::remove(const PoolId& id) {
   pools_.erase(id);
 }
 
-void* LocalPools::lookup(const PoolId& id, bool* is_device, int* device) {
+void* LocalPools::lookup(const PoolId& id, bool* is_device, int* device,
+                         uint64_t* size) {
   std::lock_guard<std::mutex> g(mu_);
   auto it = pools_.find(id);
   if (it == pools_.end()) return nullptr;
   if (is_device) *is_device = it->second.is_device;
   if (device) *device = it->second.device;
+  if (size) *size = it->second.size;
   return it->second.base;
 }
 

@@ -64,16 +64,26 @@ Result<void> TransferEngine::pull_one(StorageBackend& dst, uint64_t dst_offset,
   const bool dst_is_gpu = dst.storage_class() == StorageClass::RAM_GPU;
   uint8_t* dst_ptr = static_cast<uint8_t*>(dst.base_ptr());
 
-  // resolve source to a pointer when one-sided access is possible
+  // Bound the destination range against the pool BEFORE any transfer: the
+  // request arrives over the network-exposed data port, and the raw-pointer
+  // paths below bypass the backend's own check_range.
+  if (src.length > dst.capacity() || dst_offset > dst.capacity() - src.length)
+    return Error{ErrorCode::INVALID_OFFSET, "pull dst range out of bounds"};
+
+  // resolve source to a pointer when one-sided access is possible; when the
+  // mapping size is known, reject out-of-range source slices the same way
   const uint8_t* src_ptr = nullptr;
   bool src_is_gpu = false;
   {
     bool is_dev = false;
-    if (void* base = LocalPools::inst().lookup(src.pool_id, &is_dev)) {
+    uint64_t src_pool_size = 0;  // 0 = unknown (IPC import)
+    if (void* base = LocalPools::inst().lookup(src.pool_id, &is_dev, nullptr,
+                                               &src_pool_size)) {
       src_ptr = static_cast<const uint8_t*>(base) + src.offset;
       src_is_gpu = is_dev;
     } else if (src.access.kind == AccessKind::SHM && !src.access.shm_name.empty()) {
-      if (void* base = mapper_->map_shm(src.access.shm_name, 0))
+      if (void* base =
+              mapper_->map_shm(src.access.shm_name, 0, &src_pool_size))
         src_ptr = static_cast<const uint8_t*>(base) + src.offset;
     } else if (src.access.kind == AccessKind::HIP_IPC &&
                !src.access.ipc_handle_hex.empty() && gpu::available()) {
@@ -83,6 +93,10 @@ Result<void> TransferEngine::pull_one(StorageBackend& dst, uint64_t dst_offset,
         src_is_gpu = true;
       }
     }
+    if (src_ptr && src_pool_size > 0 &&
+        (src.length > src_pool_size ||
+         src.offset > src_pool_size - src.length))
+      return Error{ErrorCode::INVALID_OFFSET, "pull src range out of bounds"};
   }
 
   if (src_ptr && !dst_ptr) {

@@ -196,6 +196,11 @@ void WorkerService::register_handlers() {
     if (!r.ok()) return r.error();
     auto* be = backend(r->pool_id);
     if (!be) return Error{ErrorCode::POOL_NOT_FOUND, r->pool_id};
+    // Bound the attacker-controlled length BEFORE allocating the response:
+    // oversized requests must not zero-fill gigabytes (or bad_alloc) only to
+    // be rejected by the backend, nor exceed the frame limit at send time.
+    if (r->length > rpc::kMaxFrame || r->length > be->capacity())
+      return Error{ErrorCode::INVALID_OFFSET, "read length out of bounds"};
     std::string out;
     out.resize(r->length);
     BB_RETURN_IF_ERROR(be->read(r->offset, out.data(), r->length));
@@ -216,10 +221,14 @@ void WorkerService::register_handlers() {
     if (!r.ok()) return r.error();
     serde::Enc e;
     e.num<uint32_t>(static_cast<uint32_t>(r->reads.size()));
+    uint64_t total = 0;
     std::string tmp;
     for (const auto& rd : r->reads) {
       auto* be = backend(rd.pool_id);
       if (!be) return Error{ErrorCode::POOL_NOT_FOUND, rd.pool_id};
+      total += rd.length;
+      if (rd.length > be->capacity() || total > rpc::kMaxFrame)
+        return Error{ErrorCode::INVALID_OFFSET, "batch read length out of bounds"};
       tmp.resize(rd.length);
       BB_RETURN_IF_ERROR(be->read(rd.offset, tmp.data(), rd.length));
       e.bytes(tmp.data(), tmp.size());

@@ -131,3 +131,104 @@ class TestHostileFrames:
             pass
         s.close()
         self._assert_alive(server)
+
+
+class TestDataPlaneHostile:
+    """Malformed/hostile requests against the worker's network-exposed data
+    port must be rejected with a clean error before any large allocation or
+    out-of-bounds memory access (ADVICE round-1 highs)."""
+
+    HDR = "<IBQH"  # u32 body_len, u8 kind, u64 id, u16 method (packed, 15 B)
+    DATA_READ = 201
+    DATA_PULL = 206
+
+    def _call(self, endpoint, method, body, timeout=10.0):
+        import socket
+        import struct
+        host, port = endpoint.rsplit(":", 1)
+        s = socket.create_connection((host, int(port)), timeout=timeout)
+        s.settimeout(timeout)
+        try:
+            s.sendall(struct.pack(self.HDR, len(body), 0, 1, method) + body)
+            hdr = b""
+            while len(hdr) < 15:
+                chunk = s.recv(15 - len(hdr))
+                if not chunk:
+                    raise ConnectionError("server closed")
+                hdr += chunk
+            body_len, kind, rid, m = struct.unpack(self.HDR, hdr)
+            resp = b""
+            while len(resp) < body_len:
+                chunk = s.recv(min(65536, body_len - len(resp)))
+                if not chunk:
+                    raise ConnectionError("short response")
+                resp += chunk
+            status = struct.unpack("<i", resp[:4])[0]
+            (msg_len,) = struct.unpack("<I", resp[4:8])
+            msg = resp[8:8 + msg_len].decode(errors="replace")
+            return status, msg, resp[8 + msg_len:]
+        finally:
+            s.close()
+
+    @staticmethod
+    def _s(txt):
+        import struct
+        b = txt.encode()
+        return struct.pack("<I", len(b)) + b
+
+    def test_data_read_huge_length_rejected(self, cluster):
+        import struct
+        pool = cluster.workers[0].pool_descriptors()[0]
+        # length = 16 GiB: must be rejected before the response buffer resize
+        body = self._s(pool.pool_id) + struct.pack("<QQ", 0, 16 << 30)
+        status, msg, _ = self._call(cluster.workers[0].data_endpoint,
+                                    self.DATA_READ, body)
+        assert status != 0
+        assert "out of bounds" in msg
+
+    def test_data_read_beyond_capacity_rejected(self, cluster):
+        import struct
+        pool = cluster.workers[0].pool_descriptors()[0]
+        body = self._s(pool.pool_id) + struct.pack("<QQ", 0, pool.size + 4096)
+        status, msg, _ = self._call(cluster.workers[0].data_endpoint,
+                                    self.DATA_READ, body)
+        assert status != 0
+
+    def test_data_pull_dst_out_of_bounds_rejected(self, cluster):
+        import struct
+        w = cluster.workers[0]
+        pool = w.pool_descriptors()[0]
+        # craft a PullReq whose dst range overflows the destination pool and
+        # whose src resolves locally (same pool) — without the bounds check
+        # this memcpy'd past the pool
+        length = 1 << 20
+        dst_offset = pool.size - 4096  # dst_offset + length >> capacity
+        body = bb.core.encode_pull_req_for_test(
+            pool.pool_id, dst_offset, length, pool.pool_id, 0, length)
+        status, msg, _ = self._call(w.data_endpoint, self.DATA_PULL, body)
+        assert status != 0
+        assert "out of bounds" in msg
+
+    def test_data_pull_src_out_of_bounds_rejected(self, cluster):
+        w = cluster.workers[0]
+        pool = w.pool_descriptors()[0]
+        # src slice overruns the (locally resolved) source pool
+        length = 1 << 20
+        body = bb.core.encode_pull_req_for_test(
+            pool.pool_id, 0, length, pool.pool_id, pool.size - 4096, length)
+        status, msg, _ = self._call(w.data_endpoint, self.DATA_PULL, body)
+        assert status != 0
+        assert "out of bounds" in msg
+
+    def test_data_pull_hostile_vector_count_is_protocol_error(self, cluster):
+        import struct
+        w = cluster.workers[0]
+        pool = w.pool_descriptors()[0]
+        # PullReq: dst_pool, dst_offset, total_len, srcs — claim 2^32-1 shard
+        # placements with no bytes behind them. The serde guard must surface a
+        # decode failure (PROTOCOL_ERROR), not a valid empty vector.
+        body = (self._s(pool.pool_id) + struct.pack("<QQ", 0, 0)
+                + struct.pack("<I", 0xFFFFFFFF))
+        status, msg, _ = self._call(w.data_endpoint, self.DATA_PULL, body)
+        assert status != 0
+        assert "bad request" in msg
