@@ -303,6 +303,45 @@ def main():
     elapsed = time.perf_counter() - t0
     elapsed = max_over_ranks(dist, elapsed)
 
+    # ---- cold regime: same workload with the placement
+    # cache OFF and in-place upserts OFF — every step pays the full control
+    # plane (batch_put_start allocation + batch_get_workers RPC + explicit
+    # removes). Reported as value_cold so regressions on the uncached path
+    # are visible in the official record. ----
+    cold_gbps = None
+    cold_steps = max(2, args.steps // 4)
+    if use_gpu:
+        cold_cfg = bb.PlacementConfig()
+        for attr in ("replication", "checksum", "preferred_class",
+                     "preferred_worker"):
+            setattr(cold_cfg, attr, getattr(cfg, attr))
+        cold_cfg.replace = False
+        gcl_cold = bb.GpuClient(lane_clients[0], DEVICE)
+        gcl_cold.init()
+        gcl_cold.set_fused_copy(not args.no_fused_copy)
+        gcl_cold.set_placement_cache(False)
+        ck = [(f"r{RANK}cold{i}", src + i * S, S) for i in range(B)]
+        cpb = bb.make_put_batch(ck)
+        cgb = bb.make_get_batch([(k, dst + i * S, S)
+                                 for i, (k, _, _) in enumerate(ck)])
+        cold_keys = [k for k, _, _ in ck]
+
+        def cold_step():
+            assert gcl_cold.batch_put_prepared(cpb, cold_cfg), "cold put"
+            assert gcl_cold.batch_get_prepared(cgb), "cold get"
+            assert bb.client_batch_remove_prepared(lane_clients[0], cpb)
+
+        cold_step()  # warmup
+        device_sync()
+        barrier(dist)
+        t0c = time.perf_counter()
+        for _ in range(cold_steps):
+            cold_step()
+        device_sync()
+        barrier(dist)
+        elapsed_c = max_over_ranks(dist, time.perf_counter() - t0c)
+        cold_gbps = 2.0 * B * S * cold_steps * max(WORLD, 1) / elapsed_c / 1e9
+
     # ---- p50 single-object get latency (untimed probe, after the region) ----
     probe_us = []
     probe_key = f"r{RANK}probe"
@@ -337,6 +376,9 @@ def main():
             "ms_per_step": round(ms_per_step, 3),
             "higher_is_better": True,
             "scaling": "weak",
+            # same workload, placement cache off + upserts off: every step
+            # pays full allocation + metadata RPCs + removes
+            "value_cold": round(cold_gbps, 3) if cold_gbps else None,
             "vs_baseline": None,  # reference publishes no numbers
             "dtype": "uint8",
             "data": "synthetic",

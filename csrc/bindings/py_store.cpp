@@ -197,6 +197,23 @@ void bind_store(py::module_& m) {
       .def("repair_object", [](KeystoneService& k, const std::string& key) {
         unwrap_void(k.repair_object(key));
       }, py::call_guard<py::gil_scoped_release>())
+      // sessionful upsert protocol (CPU-testable surface)
+      .def("create_put_session", [](KeystoneService& k,
+                                    const std::vector<std::string>& keys,
+                                    uint64_t size, const PlacementConfig& cfg) {
+        std::vector<PutStartRequest> reqs;
+        reqs.reserve(keys.size());
+        for (auto& key : keys) reqs.push_back({key, size, cfg});
+        return k.create_put_session(reqs);
+      })
+      .def("upsert_start_token", [](KeystoneService& k, uint64_t token) {
+        unwrap_void(k.upsert_start_token(token));
+      })
+      .def("commit_token", [](KeystoneService& k, uint64_t token,
+                              const std::vector<uint64_t>& digests) {
+        unwrap_void(k.commit_token(token, digests));
+      })
+      .def("token_commits", &KeystoneService::token_commits)
       .def("coord", &KeystoneService::coord);
 
   py::class_<KeystoneServer, std::shared_ptr<KeystoneServer>>(m, "KeystoneServer")
@@ -376,9 +393,11 @@ void bind_store(py::module_& m) {
   struct DevPutBatch {
     std::vector<GpuClient::DevPutItem> items;
     std::vector<std::string> keys;
+    GpuClient::BatchPutSession session;  // steady-state token fast path
   };
   struct DevGetBatch {
     std::vector<GpuClient::DevGetItem> items;
+    GpuClient::BatchGetSession session;
   };
   py::class_<DevPutBatch>(m, "DevPutBatch")
       .def_property_readonly("size", [](const DevPutBatch& b) { return b.items.size(); });
@@ -441,21 +460,23 @@ void bind_store(py::module_& m) {
              return unwrap(g.batch_get_device(its, verify));
            }, py::arg("items"), py::arg("verify") = false)
       .def("batch_put_prepared",
-           [](GpuClient& g, const DevPutBatch& b, const PlacementConfig& cfg) {
+           [](GpuClient& g, DevPutBatch& b, const PlacementConfig& cfg) {
              py::gil_scoped_release rel;
-             auto st = unwrap(g.batch_put_device(b.items, cfg));
+             auto st = unwrap(g.batch_put_device(b.items, cfg, &b.session));
              for (auto v : st)
                if (v != 0) return false;
              return true;
            }, py::arg("batch"), py::arg("config") = PlacementConfig{})
       .def("batch_get_prepared",
-           [](GpuClient& g, const DevGetBatch& b, bool verify) {
+           [](GpuClient& g, DevGetBatch& b, bool verify) {
              py::gil_scoped_release rel;
-             auto st = unwrap(g.batch_get_device(b.items, verify));
+             auto st = unwrap(g.batch_get_device(b.items, verify, &b.session));
              for (auto v : st)
                if (v != 0) return false;
              return true;
            }, py::arg("batch"), py::arg("verify") = false)
+      .def_property_readonly("session_put_steps", &GpuClient::session_put_steps)
+      .def_property_readonly("session_get_steps", &GpuClient::session_get_steps)
       .def("batch_put_async",
            [](GpuClient& g,
               const std::vector<std::tuple<std::string, uint64_t, uint64_t>>& items,

@@ -242,9 +242,76 @@ struct PoolRef {
 };
 }  // namespace
 
+// Token fast path: placements unchanged since last step ⇒ two tiny RPCs
+// bracket ONE fused copy+digest kernel launch. Returns nullopt when the
+// session is not usable (caller runs the full path, which re-establishes it).
+std::optional<Result<std::vector<int32_t>>> GpuClient::try_session_put(
+    const std::vector<DevPutItem>& items, BatchPutSession* sess) {
+  if (!sess || sess->token == 0 || sess->owner != this ||
+      sess->descs.size() != items.size() || items.empty())
+    return std::nullopt;
+  {
+    std::lock_guard<std::mutex> g(cache_mu_);
+    if (!placement_cache_on_ || sess->cache_epoch != cache_epoch_) {
+      sess->token = 0;
+      return std::nullopt;
+    }
+  }
+  // the session is bound to one item list: same buffers, same order
+  for (size_t i = 0; i < items.size(); ++i)
+    if (sess->descs[i].src != items[i].ptr ||
+        sess->descs[i].nbytes != items[i].size) {
+      sess->token = 0;
+      return std::nullopt;
+    }
+  BB_TRACE_SCOPE("bb::session_put");
+  // RPC 1 (8 bytes): flip the session's objects to PENDING — placements are
+  // now pinned (tiering/eviction/repair only touch COMMITTED objects), so
+  // the one-sided writes below cannot race a migration
+  serde::Enc e1;
+  e1.num<uint64_t>(sess->token);
+  auto r1 = c_.meta_call_raw(M::BATCH_UPSERT_START, e1.buf);
+  if (!r1.ok()) {
+    sess->token = 0;  // stale/error before any write: clean fallback
+    return std::nullopt;
+  }
+  std::vector<uint64_t> digests(items.size(), 0);
+  auto rk = gpu::fused_put(sess->descs.data(),
+                           static_cast<uint32_t>(sess->descs.size()),
+                           digests.data(), streams_[2]);
+  if (!rk.ok()) return {rk.error()};  // objects stay PENDING; GC reclaims
+  serde::Enc e2;
+  e2.num<uint64_t>(sess->token);
+  e2.num<uint32_t>(static_cast<uint32_t>(digests.size()));
+  for (uint64_t dg : digests) e2.num<uint64_t>(dg);
+  auto r2 = c_.meta_call_raw(M::BATCH_COMMIT_TOKEN, e2.buf);
+  if (!r2.ok()) {
+    // placements changed mid-step (rare): fall back — the full path
+    // re-places and rewrites the batch
+    sess->token = 0;
+    return std::nullopt;
+  }
+  {
+    // refresh the cached digests so the verified get path accepts the new
+    // contents (raw pointers deliberately dereferenced only under the lock
+    // with the epoch re-validated)
+    std::lock_guard<std::mutex> g(cache_mu_);
+    if (sess->cache_epoch == cache_epoch_)
+      for (size_t j = 0; j < sess->entries.size(); ++j)
+        static_cast<CachedPlacement*>(sess->entries[j])->digest = digests[j];
+  }
+  session_put_steps_.fetch_add(1);
+  return {Result<std::vector<int32_t>>(std::vector<int32_t>(items.size(), 0))};
+}
+
 Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
-    const std::vector<DevPutItem>& items, const PlacementConfig& cfg) {
+    const std::vector<DevPutItem>& items, const PlacementConfig& cfg,
+    BatchPutSession* sess) {
+  if (auto fast = try_session_put(items, sess)) return std::move(*fast);
   BB_TRACE_SCOPE("bb::batch_put");
+  const bool want_token = sess != nullptr && cfg.replace &&
+                          cfg.replication <= 1 && cfg.checksum &&
+                          placement_cache_on_ && fused_copy_;
   serde::Enc req;
   req.num<uint32_t>(static_cast<uint32_t>(items.size()));
   // uniform size when possible (the common batched pattern)
@@ -256,11 +323,13 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
     for (auto& it : items) req.num<uint64_t>(it.size);
   for (auto& it : items) req.str(it.key);
   serde::put(req, cfg);
+  req.num<uint8_t>(want_token ? 1 : 0);
   auto resp = c_.meta_call_raw(rpc::methods::BATCH_PUT_START2, req.buf);
   if (!resp.ok()) return resp.error();
 
   serde::Dec d(resp.value().data(), resp.value().size());
   d.num<uint64_t>();  // view version
+  const uint64_t token = d.num<uint64_t>();  // 0 = no session granted
   const uint16_t npools = d.num<uint16_t>();
   std::vector<PoolRef> pools(npools);
   for (uint16_t i = 0; i < npools; ++i) {
@@ -398,7 +467,32 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
                                   fused_hash_loc[j].second, it.size,
                                   fused_digests[j]};
     }
-    if (placement_cache_.size() > (1u << 20)) placement_cache_.clear();
+    if (placement_cache_.size() > (1u << 20)) {
+      placement_cache_.clear();
+      ++cache_epoch_;
+    }
+    // establish the batch session when the server granted a token AND every
+    // item rode the fused single-copy path (so desc j == item j)
+    if (sess && token != 0 && fused_hash_idx.size() == items.size()) {
+      bool all_cached = true;
+      sess->descs = fused_hash;
+      sess->entries.clear();
+      sess->entries.reserve(items.size());
+      for (size_t j = 0; j < items.size() && all_cached; ++j) {
+        auto cit = placement_cache_.find(items[fused_hash_idx[j]].key);
+        if (cit == placement_cache_.end()) all_cached = false;
+        else sess->entries.push_back(&cit->second);
+      }
+      if (all_cached) {
+        sess->token = token;
+        sess->cache_epoch = cache_epoch_;
+        sess->owner = this;
+      } else {
+        sess->token = 0;
+        sess->descs.clear();
+        sess->entries.clear();
+      }
+    }
   }
   std::vector<std::string> cancels;
   for (size_t i = 0; i < items.size(); ++i)
@@ -415,16 +509,19 @@ void GpuClient::set_placement_cache(bool on) {
   std::lock_guard<std::mutex> g(cache_mu_);
   placement_cache_on_ = on;
   if (!on) placement_cache_.clear();
+  ++cache_epoch_;
 }
 
 void GpuClient::invalidate(const std::vector<ObjectKey>& keys) {
   std::lock_guard<std::mutex> g(cache_mu_);
   for (const auto& k : keys) placement_cache_.erase(k);
+  ++cache_epoch_;
 }
 
 void GpuClient::clear_placement_cache() {
   std::lock_guard<std::mutex> g(cache_mu_);
   placement_cache_.clear();
+  ++cache_epoch_;
 }
 
 uint8_t* GpuClient::device_pool_base(const PoolId& id) {
@@ -440,8 +537,63 @@ uint8_t* GpuClient::device_pool_base(const PoolId& id) {
   return nullptr;
 }
 
+// Session fast path for gets: descs and want-digest slots were resolved on a
+// previous step; a step is ONE kernel launch + digest compares, zero RPCs.
+std::optional<Result<std::vector<int32_t>>> GpuClient::try_session_get(
+    const std::vector<DevGetItem>& items, BatchGetSession* sess) {
+  if (!sess || !sess->complete || sess->owner != this ||
+      sess->descs.size() != items.size() || items.empty())
+    return std::nullopt;
+  std::vector<uint64_t> want(items.size());
+  {
+    std::lock_guard<std::mutex> g(cache_mu_);
+    if (!placement_cache_on_ || sess->cache_epoch != cache_epoch_) {
+      sess->complete = false;
+      return std::nullopt;
+    }
+    for (size_t j = 0; j < items.size(); ++j)
+      want[j] = static_cast<CachedPlacement*>(sess->entries[j])->digest;
+  }
+  for (size_t i = 0; i < items.size(); ++i)
+    if (sess->descs[i].dst != items[i].ptr ||
+        sess->descs[i].nbytes > items[i].capacity) {
+      sess->complete = false;
+      return std::nullopt;
+    }
+  BB_TRACE_SCOPE("bb::session_get");
+  std::vector<uint64_t> got(items.size(), 0);
+  auto rk = gpu::fused_put(sess->descs.data(),
+                           static_cast<uint32_t>(sess->descs.size()),
+                           got.data(), streams_[2]);
+  if (!rk.ok()) return {rk.error()};
+  std::vector<uint32_t> miss;
+  for (size_t j = 0; j < items.size(); ++j)
+    if (got[j] != want[j]) miss.push_back(static_cast<uint32_t>(j));
+  if (miss.empty()) {
+    session_get_steps_.fetch_add(1);
+    return {Result<std::vector<int32_t>>(std::vector<int32_t>(items.size(), 0))};
+  }
+  // stale entries: drop them (epoch bump invalidates sessions) and refetch
+  // the misses authoritatively
+  {
+    std::lock_guard<std::mutex> g(cache_mu_);
+    for (auto j : miss) placement_cache_.erase(items[j].key);
+    ++cache_epoch_;
+  }
+  sess->complete = false;
+  std::vector<int32_t> statuses(items.size(), 0);
+  std::vector<DevGetItem> sub;
+  sub.reserve(miss.size());
+  for (auto j : miss) sub.push_back(items[j]);
+  auto r = batch_get_device_rpc(sub, /*verify=*/true);
+  if (!r.ok()) return {r.error()};
+  for (size_t j = 0; j < miss.size(); ++j) statuses[miss[j]] = r.value()[j];
+  return {Result<std::vector<int32_t>>(std::move(statuses))};
+}
+
 Result<std::vector<int32_t>> GpuClient::batch_get_device_v2(
-    const std::vector<DevGetItem>& items, bool verify) {
+    const std::vector<DevGetItem>& items, bool verify, BatchGetSession* sess) {
+  if (auto fast = try_session_get(items, sess)) return std::move(*fast);
   bool cache_on;
   {
     std::lock_guard<std::mutex> g(cache_mu_);
@@ -457,18 +609,25 @@ Result<std::vector<int32_t>> GpuClient::batch_get_device_v2(
   std::vector<uint64_t> want;
   std::vector<uint32_t> miss_idx;
   std::vector<std::pair<uint32_t, CachedPlacement>> lookups;
+  std::vector<CachedPlacement*> entry_ptrs;  // parallel to lookups
+  uint64_t epoch_at_lookup = 0;
   {
     std::lock_guard<std::mutex> g(cache_mu_);
     for (size_t i = 0; i < items.size(); ++i) {
       auto it = placement_cache_.find(items[i].key);
-      if (it != placement_cache_.end() && it->second.size <= items[i].capacity)
+      if (it != placement_cache_.end() && it->second.size <= items[i].capacity) {
         lookups.emplace_back(static_cast<uint32_t>(i), it->second);
-      else
+        entry_ptrs.push_back(&it->second);
+      } else {
         miss_idx.push_back(static_cast<uint32_t>(i));
+      }
     }
+    epoch_at_lookup = cache_epoch_;
   }
   // pool resolution may RPC (view-versioned pool cache) — outside the lock
-  for (auto& [i, cp] : lookups) {
+  std::vector<CachedPlacement*> hit_entries;
+  for (size_t li = 0; li < lookups.size(); ++li) {
+    auto& [i, cp] = lookups[li];
     uint8_t* base = device_pool_base(cp.pool_id);
     const auto du = reinterpret_cast<uintptr_t>(items[i].ptr);
     if (base &&
@@ -477,11 +636,13 @@ Result<std::vector<int32_t>> GpuClient::batch_get_device_v2(
           {base + cp.offset, static_cast<uint8_t*>(items[i].ptr), cp.size});
       hit_idx.push_back(i);
       want.push_back(cp.digest);
+      hit_entries.push_back(entry_ptrs[li]);
     } else {
       miss_idx.push_back(i);
     }
   }
   std::vector<int32_t> statuses(items.size(), 0);
+  bool all_verified = false;
   if (!descs.empty()) {
     BB_TRACE_SCOPE("bb::cached_get");
     std::vector<uint64_t> got(descs.size(), 0);
@@ -489,11 +650,24 @@ Result<std::vector<int32_t>> GpuClient::batch_get_device_v2(
                             got.data(), streams_[2]);
     if (!r.ok()) return r.error();
     std::lock_guard<std::mutex> g(cache_mu_);
+    all_verified = true;
     for (size_t j = 0; j < hit_idx.size(); ++j) {
       if (got[j] != want[j]) {
         placement_cache_.erase(items[hit_idx[j]].key);
+        ++cache_epoch_;
         miss_idx.push_back(hit_idx[j]);  // refetch authoritatively
+        all_verified = false;
       }
+    }
+    // establish the get session when every item was a verified cache hit
+    // (desc j == item j) and no concurrent thread mutated the cache
+    if (sess && all_verified && miss_idx.empty() &&
+        hit_idx.size() == items.size() && epoch_at_lookup == cache_epoch_) {
+      sess->descs = std::move(descs);
+      sess->entries.assign(hit_entries.begin(), hit_entries.end());
+      sess->cache_epoch = cache_epoch_;
+      sess->owner = this;
+      sess->complete = true;
     }
   }
   if (!miss_idx.empty()) {
@@ -668,10 +842,12 @@ Result<std::vector<int32_t>> GpuClient::batch_get_device_rpc(
 // -------------------------------------------------------------- batch ops
 
 Result<std::vector<int32_t>> GpuClient::batch_put_device(
-    const std::vector<DevPutItem>& items, const PlacementConfig& cfg) {
+    const std::vector<DevPutItem>& items, const PlacementConfig& cfg,
+    BatchPutSession* sess) {
   BB_RETURN_IF_ERROR(init());
   BB_HIP(hipSetDevice(device_));
-  if (cfg.max_workers_per_copy <= 1) return batch_put_device_v2(items, cfg);
+  if (cfg.max_workers_per_copy <= 1)
+    return batch_put_device_v2(items, cfg, sess);
 
   BatchPutStartRequest breq;
   breq.requests.reserve(items.size());
@@ -794,11 +970,11 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device(
 }
 
 Result<std::vector<int32_t>> GpuClient::batch_get_device(
-    const std::vector<DevGetItem>& items, bool verify) {
+    const std::vector<DevGetItem>& items, bool verify, BatchGetSession* sess) {
   BB_RETURN_IF_ERROR(init());
   BB_HIP(hipSetDevice(device_));
   {
-    auto v2 = batch_get_device_v2(items, verify);
+    auto v2 = batch_get_device_v2(items, verify, sess);
     // a striped object in the batch makes the server signal fallback per
     // item; only a whole-response failure falls back to v1
     if (v2.ok()) {

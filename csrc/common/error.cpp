@@ -38,6 +38,7 @@ std::string_view to_string(ErrorCode c) {
     case ErrorCode::CHECKSUM_MISMATCH: return "CHECKSUM_MISMATCH";
     case ErrorCode::OBJECT_NOT_COMMITTED: return "OBJECT_NOT_COMMITTED";
     case ErrorCode::NO_PLACEMENT: return "NO_PLACEMENT";
+    case ErrorCode::SESSION_STALE: return "SESSION_STALE";
     case ErrorCode::INVALID_ARGUMENT: return "INVALID_ARGUMENT";
     case ErrorCode::TRANSFER_FAILED: return "TRANSFER_FAILED";
     case ErrorCode::NOT_CONNECTED: return "NOT_CONNECTED";

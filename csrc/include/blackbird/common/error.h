@@ -56,6 +56,7 @@ enum class ErrorCode : int32_t {
   CHECKSUM_MISMATCH = 5003,
   OBJECT_NOT_COMMITTED = 5004,
   NO_PLACEMENT = 5005,
+  SESSION_STALE = 5006,  // put-session token no longer matches server state
 
   // CLIENT (6000)
   INVALID_ARGUMENT = 6000,

@@ -62,6 +62,21 @@ class KeystoneService {
   std::vector<uint8_t> batch_object_exists(const std::vector<ObjectKey>& keys);
   std::vector<int32_t> batch_remove(const std::vector<ObjectKey>& keys);
 
+  // ---- sessionful upserts (steady-state small-object fast path) ----
+  // A put session pins an ordered key list server-side so repeat puts of the
+  // same batch cost two tiny RPCs: upsert_start_token (flip every object to
+  // PENDING, which pins placements against tiering/eviction) and
+  // commit_token (record digests, flip back to COMMITTED). Any placement
+  // change since session creation (remove / migration / scrub / repair /
+  // worker death) bumps placement_epoch_ and the session fails with
+  // SESSION_STALE — the client falls back to the full batch_put_start path.
+  // No UCX analogue in the reference; its closest pattern is the batch RPC
+  // loop at keystone_service.cpp:302-360.
+  uint64_t create_put_session(const std::vector<PutStartRequest>& reqs);
+  Result<void> upsert_start_token(uint64_t token);
+  Result<void> commit_token(uint64_t token, const std::vector<uint64_t>& digests);
+  uint64_t token_commits() const { return ctr_token_commits_.load(); }
+
   // ------------------------------------------------------ cluster view
   std::vector<WorkerInfo> get_workers_info();
   std::vector<MemoryPool> get_memory_pools();
@@ -143,6 +158,24 @@ class KeystoneService {
 
   std::shared_mutex objects_mu_;
   std::unordered_map<ObjectKey, ObjectMeta> objects_;
+
+  // ---- put sessions ----
+  // metas are raw pointers into objects_ (unordered_map mapped values are
+  // stable across insert/rehash); every objects_.erase or copies mutation
+  // bumps placement_epoch_ under objects_mu_, so a session whose epoch no
+  // longer matches is never dereferenced.
+  struct PutSession {
+    std::vector<ObjectMeta*> metas;
+    std::vector<uint64_t> sizes;
+    uint64_t epoch = 0;
+    uint64_t created_ms = 0;
+  };
+  uint64_t placement_epoch_ = 0;  // guarded by objects_mu_ (unique)
+  void bump_placement_epoch_locked() { ++placement_epoch_; }
+  std::mutex sessions_mu_;
+  std::unordered_map<uint64_t, std::shared_ptr<PutSession>> put_sessions_;
+  std::atomic<uint64_t> next_session_token_{1};
+  std::atomic<uint64_t> ctr_token_commits_{0};
 
   std::shared_mutex workers_mu_;
   std::map<WorkerId, WorkerInfo> workers_;

@@ -36,6 +36,13 @@ constexpr uint16_t ADMIN_SCRUB = 23;
 constexpr uint16_t ADMIN_REPAIR = 24;
 constexpr uint16_t ADMIN_COMPACT = 25;
 
+// sessionful upsert protocol: a BATCH_PUT_START2 response can carry a
+// session token; steady-state re-puts of the same batch then cost two tiny
+// RPCs (start: 8-byte token; commit: token + per-item digests) instead of
+// re-sending every key both ways
+constexpr uint16_t BATCH_UPSERT_START = 26;
+constexpr uint16_t BATCH_COMMIT_TOKEN = 27;
+
 // worker data plane (TCP fallback path; SHM/HIP-IPC paths bypass RPC)
 constexpr uint16_t DATA_WRITE = 200;
 constexpr uint16_t DATA_READ = 201;

@@ -189,14 +189,24 @@ void KeystoneServer::register_handlers() {
       for (uint32_t i = 0; i < count; ++i) sizes[i] = d.num<uint64_t>();
     for (uint32_t i = 0; i < count; ++i) keys[i] = d.str();
     serde::get(d, cfg);
+    const uint8_t want_token = d.remaining() ? d.num<uint8_t>() : 0;
     if (!d.ok()) return Error{ErrorCode::PROTOCOL_ERROR, "bad v2 request"};
     cfg.max_workers_per_copy = 1;  // v2 contract: single-shard copies
     for (uint32_t i = 0; i < count; ++i)
       reqs.push_back({std::move(keys[i]), sizes[i], cfg});
     auto resp = ks.batch_put_start(reqs);
 
+    uint64_t token = 0;
+    if (want_token) {
+      bool all_ok = !resp.items.empty();
+      for (auto& it : resp.items)
+        if (it.status != 0) { all_ok = false; break; }
+      if (all_ok) token = ks.create_put_session(reqs);
+    }
+
     serde::Enc e;
     e.num<uint64_t>(resp.view_version);
+    e.num<uint64_t>(token);  // 0 = no session granted
     // build the pool table
     std::map<std::string, uint16_t> pool_idx;
     std::vector<const std::string*> table;
@@ -262,6 +272,24 @@ void KeystoneServer::register_handlers() {
       }
     }
     return std::move(e.buf);
+  });
+  rpc_.register_handler(M::BATCH_UPSERT_START, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
+    serde::Dec d(b.data(), b.size());
+    uint64_t token = d.num<uint64_t>();
+    if (!d.ok()) return Error{ErrorCode::PROTOCOL_ERROR, "bad upsert request"};
+    BB_RETURN_IF_ERROR(ks.upsert_start_token(token));
+    return std::string{};
+  });
+  rpc_.register_handler(M::BATCH_COMMIT_TOKEN, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
+    serde::Dec d(b.data(), b.size());
+    uint64_t token = d.num<uint64_t>();
+    uint32_t n = d.num<uint32_t>();
+    if (!d.ok() || d.remaining() != n * sizeof(uint64_t) || n > (1u << 24))
+      return Error{ErrorCode::PROTOCOL_ERROR, "bad commit request"};
+    std::vector<uint64_t> digests(n);
+    for (uint32_t i = 0; i < n; ++i) digests[i] = d.num<uint64_t>();
+    BB_RETURN_IF_ERROR(ks.commit_token(token, digests));
+    return std::string{};
   });
   rpc_.register_handler(M::BATCH_REMOVE, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
     auto r = decode<KeysMsg>(b);

@@ -193,6 +193,7 @@ Result<void> KeystoneService::put_cancel(const ObjectKey& key) {
 Result<void> KeystoneService::remove_object_locked(const ObjectKey& key) {
   allocator_.free(key);
   objects_.erase(key);
+  bump_placement_epoch_locked();  // session meta pointers may now dangle
   mark_dirty_locked(key, true);
   bump_view();
   return {};
@@ -235,6 +236,7 @@ uint64_t KeystoneService::remove_all_objects() {
   uint64_t n = objects_.size();
   for (auto& [key, meta] : objects_) allocator_.free(key);
   objects_.clear();
+  bump_placement_epoch_locked();
   bump_view();
   return n;
 }
@@ -315,6 +317,7 @@ BatchPutStartResponse KeystoneService::batch_put_start(
         // the ranges in one allocator batch below (not per key under this
         // lock)
         objects_.erase(it);
+        bump_placement_epoch_locked();
         mark_dirty_locked(reqs[i].key, true);
         replaced.push_back(&reqs[i].key);
       }
@@ -400,6 +403,107 @@ std::vector<int32_t> KeystoneService::batch_put_complete(
   return out;
 }
 
+// ------------------------------------------------- sessionful upserts
+
+uint64_t KeystoneService::create_put_session(
+    const std::vector<PutStartRequest>& reqs) {
+  if (reqs.empty()) return 0;
+  auto s = std::make_shared<PutSession>();
+  s->metas.reserve(reqs.size());
+  s->sizes.reserve(reqs.size());
+  s->created_ms = now_ms();
+  {
+    std::shared_lock lk(objects_mu_);
+    for (const auto& r : reqs) {
+      auto it = objects_.find(r.key);
+      // sessions only cover single-copy single-shard objects: the token
+      // commit path records ONE digest per object, and the client's fused
+      // write path resolves exactly one destination
+      if (it == objects_.end() || it->second.copies.size() != 1 ||
+          it->second.copies[0].shards.size() != 1)
+        return 0;
+      s->metas.push_back(&it->second);
+      s->sizes.push_back(it->second.size);
+    }
+    s->epoch = placement_epoch_;
+  }
+  uint64_t token = next_session_token_.fetch_add(1);
+  std::lock_guard<std::mutex> g(sessions_mu_);
+  put_sessions_[token] = std::move(s);
+  if (put_sessions_.size() > 1024) {
+    // cap the table: evict the oldest session (tokens are monotonic)
+    auto oldest = put_sessions_.begin();
+    for (auto it = put_sessions_.begin(); it != put_sessions_.end(); ++it)
+      if (it->first < oldest->first) oldest = it;
+    put_sessions_.erase(oldest);
+  }
+  return token;
+}
+
+Result<void> KeystoneService::upsert_start_token(uint64_t token) {
+  if (!is_leader())
+    return Error{ErrorCode::NOT_LEADER, "standby keystone; retry on leader"};
+  std::shared_ptr<PutSession> s;
+  {
+    std::lock_guard<std::mutex> g(sessions_mu_);
+    auto it = put_sessions_.find(token);
+    if (it == put_sessions_.end())
+      return Error{ErrorCode::SESSION_STALE, "unknown put session"};
+    s = it->second;
+  }
+  std::unique_lock lk(objects_mu_);
+  if (s->epoch != placement_epoch_)
+    return Error{ErrorCode::SESSION_STALE, "placements changed"};
+  // validate ALL before flipping ANY (all-or-nothing)
+  for (size_t i = 0; i < s->metas.size(); ++i) {
+    const ObjectMeta* m = s->metas[i];
+    // PENDING allowed: a client that died between start and commit may
+    // retry the same session
+    if (m->size != s->sizes[i])
+      return Error{ErrorCode::SESSION_STALE, "object shape changed"};
+  }
+  // PENDING pins the placements: tiering/eviction/repair/scrub only touch
+  // COMMITTED objects, so the client's one-sided writes land in ranges that
+  // cannot move underneath them
+  for (auto* m : s->metas) m->state = ObjectState::PENDING;
+  return {};
+}
+
+Result<void> KeystoneService::commit_token(
+    uint64_t token, const std::vector<uint64_t>& digests) {
+  if (!is_leader())
+    return Error{ErrorCode::NOT_LEADER, "standby keystone; retry on leader"};
+  std::shared_ptr<PutSession> s;
+  {
+    std::lock_guard<std::mutex> g(sessions_mu_);
+    auto it = put_sessions_.find(token);
+    if (it == put_sessions_.end())
+      return Error{ErrorCode::SESSION_STALE, "unknown put session"};
+    s = it->second;
+  }
+  if (digests.size() != s->metas.size())
+    return Error{ErrorCode::INVALID_ARGUMENT, "digest count mismatch"};
+  const uint64_t now = now_ms();
+  std::unique_lock lk(objects_mu_);
+  if (s->epoch != placement_epoch_)
+    return Error{ErrorCode::SESSION_STALE, "placements changed"};
+  for (size_t i = 0; i < s->metas.size(); ++i)
+    if (s->metas[i]->size != s->sizes[i])
+      return Error{ErrorCode::SESSION_STALE, "object shape changed"};
+  const bool persist = config_.persist_objects;
+  for (size_t i = 0; i < s->metas.size(); ++i) {
+    ObjectMeta* m = s->metas[i];
+    m->state = ObjectState::COMMITTED;
+    m->checksum = digests[i];
+    m->created_ms = now;  // TTL restarts at commit, as in put_complete
+    m->last_access_ms = now;
+    if (persist) mark_dirty_locked(m->key, false);
+  }
+  bump_view();
+  ctr_token_commits_.fetch_add(1);
+  return {};
+}
+
 std::vector<int32_t> KeystoneService::batch_put_cancel(
     const std::vector<ObjectKey>& keys) {
   std::vector<int32_t> out;
@@ -465,6 +569,7 @@ std::vector<int32_t> KeystoneService::batch_remove(
         continue;
       }
       objects_.erase(it);
+      bump_placement_epoch_locked();
       mark_dirty_locked(keys[i], true);
       to_free.push_back(&keys[i]);
     }
@@ -754,6 +859,7 @@ Result<void> KeystoneService::migrate_object(const ObjectKey& key,
     }
     it->second.copies = std::move(placed.value());
     it->second.access_count = 0;
+    bump_placement_epoch_locked();
     mark_dirty_locked(key, false);
     bump_view();
   }
@@ -974,6 +1080,7 @@ Result<void> KeystoneService::repair_object(const ObjectKey& key) {
       return mr.error();
     }
     it->second.copies.push_back(std::move(placed.value()));
+    bump_placement_epoch_locked();
     mark_dirty_locked(key, false);
     bump_view();
   }
@@ -1190,6 +1297,7 @@ uint32_t KeystoneService::run_scrub_once(uint32_t max_objects) {
         for (auto cit = copies.begin(); cit != copies.end(); ++cit) {
           if (cit->shards == shards) {
             copies.erase(cit);
+            bump_placement_epoch_locked();
             dropped = true;
             break;
           }
@@ -1380,6 +1488,7 @@ void KeystoneService::cleanup_dead_worker(const WorkerId& id) {
       // (double allocation / freeing live ranges).
       for (auto it = dead; it != copies.end(); ++it)
         allocator_.free_ranges(key, it->shards);
+      if (dead != copies.end()) bump_placement_epoch_locked();
       copies.erase(dead, copies.end());
       if (copies.empty() && before > 0) lost.push_back(key);
     }

@@ -629,3 +629,89 @@ class TestGpuStriping:
             c.close()
         finally:
             cl.stop()
+
+
+class TestBatchSessions:
+    def test_session_fast_path_roundtrip(self):
+        """Prepared batches establish a put session on the first step; later
+        steps ride the token fast path (two tiny RPCs, one fused kernel) and
+        gets go fully RPC-free. Data must stay correct as the source buffer
+        changes between steps, and server-side interference must invalidate
+        the session transparently."""
+        cl = Cluster(n_workers=1, pool_bytes=256 * MB,
+                     storage_class=bb.StorageClass.RAM_GPU)
+        g = bb.core.gpu
+        try:
+            c = cl.client()
+            gcl = bb.GpuClient(c, 0)
+            gcl.init()
+            gcl.set_placement_cache(True)
+            N, S = 32, 64 * 1024
+            src = g.malloc(N * S)
+            dst = g.malloc(N * S)
+            cfg = bb.PlacementConfig()
+            cfg.replace = True
+            pb = bb.make_put_batch([("sess%02d" % i, src + i * S, S)
+                                    for i in range(N)])
+            gb = bb.make_get_batch([("sess%02d" % i, dst + i * S, S)
+                                    for i in range(N)])
+            ks = cl.keystone.service()
+
+            for step in range(4):
+                blobs = [os.urandom(S) for _ in range(N)]
+                for i, b in enumerate(blobs):
+                    g.upload(src + i * S, b)
+                assert gcl.batch_put_prepared(pb, cfg)
+                assert gcl.batch_get_prepared(gb)
+                for i, b in enumerate(blobs):
+                    assert g.download(dst + i * S, S) == b, (step, i)
+                # keystone's authoritative digests match the CPU reference
+                info = ks.get_workers("sess07")
+                assert info.checksum == g.checksum_cpu(blobs[7])
+            # steps 2..4 must have used the token fast path and RPC-free gets
+            assert gcl.session_put_steps >= 3, gcl.session_put_steps
+            assert gcl.session_get_steps >= 2, gcl.session_get_steps
+            assert ks.token_commits() >= 3
+
+            # server-side interference: removing ANY object bumps the
+            # placement epoch → session falls back transparently, still OK
+            ks.put_start("intruder", 4096, bb.PlacementConfig())
+            ks.put_complete("intruder", checksum=1)
+            ks.remove_object("intruder")
+            blobs = [os.urandom(S) for _ in range(N)]
+            for i, b in enumerate(blobs):
+                g.upload(src + i * S, b)
+            fast_before = gcl.session_put_steps
+            assert gcl.batch_put_prepared(pb, cfg)   # full path this step
+            assert gcl.batch_put_prepared(pb, cfg)   # session re-established
+            assert gcl.session_put_steps == fast_before + 1
+            assert gcl.batch_get_prepared(gb)
+            for i, b in enumerate(blobs):
+                assert g.download(dst + i * S, S) == b, i
+            c.close()
+            g.free(src)
+            g.free(dst)
+        finally:
+            cl.stop()
+
+    def test_session_not_used_when_replicated_or_cache_off(self):
+        cl = Cluster(n_workers=1, pool_bytes=256 * MB,
+                     storage_class=bb.StorageClass.RAM_GPU)
+        g = bb.core.gpu
+        try:
+            c = cl.client()
+            gcl = bb.GpuClient(c, 0)
+            gcl.init()  # placement cache OFF
+            N, S = 4, 4096
+            src = g.malloc(N * S)
+            cfg = bb.PlacementConfig()
+            cfg.replace = True
+            pb = bb.make_put_batch([("nc%d" % i, src + i * S, S)
+                                    for i in range(N)])
+            for _ in range(3):
+                assert gcl.batch_put_prepared(pb, cfg)
+            assert gcl.session_put_steps == 0
+            g.free(src)
+            c.close()
+        finally:
+            cl.stop()

@@ -426,3 +426,87 @@ class TestHighAvailability:
             k2.stop()
             k2.service().stop()
             cs.stop()
+
+
+class TestPutSessions:
+    """Sessionful upsert protocol: commit-by-token instead of re-sending keys
+    (the steady-state small-object fast path)."""
+
+    def _put(self, ks, keys, size=4096, checksum=7):
+        cfg = bb.PlacementConfig()
+        for k in keys:
+            ks.put_start(k, size, cfg)
+            ks.put_complete(k, checksum=checksum)
+
+    def test_token_roundtrip(self, ks):
+        keys = ["s%d" % i for i in range(16)]
+        self._put(ks, keys)
+        tok = ks.create_put_session(keys, 4096, bb.PlacementConfig())
+        assert tok != 0
+        before = ks.token_commits()
+        ks.upsert_start_token(tok)
+        # PENDING pins: objects invisible to readers during the write window
+        with pytest.raises(Exception, match="OBJECT_NOT_COMMITTED"):
+            ks.get_workers(keys[0])
+        ks.commit_token(tok, [100 + i for i in range(16)])
+        assert ks.token_commits() == before + 1
+        for i, k in enumerate(keys):
+            info = ks.get_workers(k)
+            assert info.checksum == 100 + i
+        # reusable across steps
+        ks.upsert_start_token(tok)
+        ks.commit_token(tok, [200 + i for i in range(16)])
+        assert ks.get_workers(keys[3]).checksum == 203
+
+    def test_unknown_token(self, ks):
+        with pytest.raises(Exception, match="SESSION_STALE"):
+            ks.upsert_start_token(999999)
+        with pytest.raises(Exception, match="SESSION_STALE"):
+            ks.commit_token(999999, [1])
+
+    def test_digest_count_mismatch(self, ks):
+        keys = ["m%d" % i for i in range(4)]
+        self._put(ks, keys)
+        tok = ks.create_put_session(keys, 4096, bb.PlacementConfig())
+        ks.upsert_start_token(tok)
+        with pytest.raises(Exception, match="INVALID_ARGUMENT"):
+            ks.commit_token(tok, [1, 2])  # 2 digests for 4 objects
+
+    def test_remove_staleness(self, ks):
+        # ANY object removal invalidates every session (placement epoch):
+        # meta pointers may dangle and ranges may be reallocated
+        keys = ["r%d" % i for i in range(4)]
+        self._put(ks, keys)
+        tok = ks.create_put_session(keys, 4096, bb.PlacementConfig())
+        ks.put_start("other", 4096, bb.PlacementConfig())
+        ks.put_complete("other", checksum=1)
+        ks.remove_object("other")
+        with pytest.raises(Exception, match="SESSION_STALE"):
+            ks.upsert_start_token(tok)
+        with pytest.raises(Exception, match="SESSION_STALE"):
+            ks.commit_token(tok, [0] * 4)
+
+    def test_session_not_granted_for_replicated(self, ks):
+        # token commits record ONE digest per object: multi-copy objects are
+        # excluded at session creation
+        ks.register_pool(make_pool("p1", worker="w1"))
+        cfg = bb.PlacementConfig()
+        cfg.replication = 2
+        ks.put_start("rep", 4096, cfg)
+        ks.put_complete("rep", checksum=5)
+        assert ks.create_put_session(["rep"], 4096, cfg) == 0
+
+    def test_session_not_granted_for_missing(self, ks):
+        assert ks.create_put_session(["nope"], 4096, bb.PlacementConfig()) == 0
+
+    def test_commit_requires_matching_shape(self, ks):
+        # replacing an object with a different size between session creation
+        # and use bumps the epoch (remove+realloc) → stale
+        self._put(ks, ["shape"], size=4096)
+        tok = ks.create_put_session(["shape"], 4096, bb.PlacementConfig())
+        cfg = bb.PlacementConfig()
+        cfg.replace = True
+        ks.put_start("shape", 8192, cfg)  # different size: remove + realloc
+        ks.put_complete("shape", checksum=9)
+        with pytest.raises(Exception, match="SESSION_STALE"):
+            ks.upsert_start_token(tok)
