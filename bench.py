@@ -342,6 +342,88 @@ def main():
         elapsed_c = max_over_ranks(dist, time.perf_counter() - t0c)
         cold_gbps = 2.0 * B * S * cold_steps * max(WORLD, 1) / elapsed_c / 1e9
 
+    # ---- multi-rank shuffle A/B: the symmetric "every rank
+    # fetches a peer's batch" pattern via (a) the one-sided IPC gather path
+    # and (b) the RCCL all-to-all batch shuffle. Untimed side-measurement;
+    # both numbers go into config so the driver's 2/4/8-GPU runs produce the
+    # comparison. ----
+    def all_ranks_ok(flag):
+        """Consensus gate: a rank that failed setup must not leave peers
+        blocked at a barrier — every timed phase starts only if ALL agree."""
+        if dist is None:
+            return flag
+        import torch
+        t = torch.tensor([1.0 if flag else 0.0], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MIN)
+        return t.item() > 0.5
+
+    shuffle_ab = None
+    if use_gpu and WORLD > 1:
+        peer = (RANK + 1) % WORLD
+        peer_keys = [f"r{peer}L0o{i}" for i in range(B)]
+        ab_steps = 5
+        recv = bb.core.gpu.malloc(B * S, DEVICE)
+        shuffle_ab = {}
+
+        # (a) one-sided IPC gather of the peer's batch (cross-GPU xGMI)
+        ipc_items = [(k, recv + i * S, S) for i, k in enumerate(peer_keys)]
+        gcl_ab = None
+        ok = True
+        try:
+            gcl_ab = bb.GpuClient(lane_clients[0], DEVICE)
+            gcl_ab.init()
+            gcl_ab.set_fused_copy(not args.no_fused_copy)
+            assert gcl_ab.batch_get_device(ipc_items) == [0] * B
+            bb.core.gpu.sync()
+        except Exception as e:
+            log(f"shuffle A/B ipc setup failed: {e}")
+            shuffle_ab["ipc_error"] = str(e)[:200]
+            ok = False
+        if all_ranks_ok(ok):
+            barrier(dist)
+            t0s = time.perf_counter()
+            for _ in range(ab_steps):
+                assert gcl_ab.batch_get_device(ipc_items) == [0] * B
+            bb.core.gpu.sync()
+            barrier(dist)
+            ipc_s = max_over_ranks(dist, time.perf_counter() - t0s)
+            shuffle_ab["ipc_gather_gbps"] = round(
+                B * S * ab_steps * WORLD / ipc_s / 1e9, 1)
+
+        # (b) RCCL all-to-all shuffle of the same batches
+        eng = coord_cl = None
+        ok = True
+        try:
+            coord_cl = bb.CoordClient()
+            coord_cl.connect(coord_ep)
+            eng = bb.RcclEngine()
+            eng.init(coord_cl, "default", f"bench-ab-{base_port}", RANK,
+                     WORLD, DEVICE)
+        except Exception as e:  # e.g. ranks sharing one device (RCCL refuses)
+            log(f"shuffle A/B rccl init failed: {e}")
+            shuffle_ab["rccl_error"] = str(e)[:200]
+            ok = False
+        if all_ranks_ok(ok):
+            wants = [([], [], 0)] * WORLD
+            wants[peer] = (peer_keys, [S] * B, recv)
+            bb.core.gpu_batch_shuffle(gcl, eng, wants)  # warmup
+            bb.core.gpu.sync()
+            barrier(dist)
+            t0s = time.perf_counter()
+            for _ in range(ab_steps):
+                bb.core.gpu_batch_shuffle(gcl, eng, wants)
+            bb.core.gpu.sync()
+            barrier(dist)
+            rccl_s = max_over_ranks(dist, time.perf_counter() - t0s)
+            shuffle_ab["rccl_alltoall_gbps"] = round(
+                B * S * ab_steps * WORLD / rccl_s / 1e9, 1)
+            log(f"shuffle A/B: {shuffle_ab}")
+        if eng is not None:
+            eng.destroy()
+        if coord_cl is not None:
+            coord_cl.close()
+        bb.core.gpu.free(recv)
+
     # ---- p50 single-object get latency (untimed probe, after the region) ----
     probe_us = []
     probe_key = f"r{RANK}probe"
@@ -395,6 +477,7 @@ def main():
                 "fused_copy": not args.no_fused_copy,
                 "pipeline_lanes": lanes,
                 "p50_get_latency_us": round(p50_us, 1),
+                "shuffle_ab": shuffle_ab,
             },
         }
         print(json.dumps(out), flush=True)

@@ -1,10 +1,14 @@
 // Bindings for the keystone / worker / client layers.
+#include <map>
+#include <thread>
+
 #include <pybind11/functional.h>
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
 #include "blackbird/client/client.h"
 #include "blackbird/client/gpu_client.h"
+#include "blackbird/client/shuffle.h"
 #include "blackbird/gpu/gpu_kernels.h"
 #include "blackbird/keystone/keystone_rpc.h"
 #include "blackbird/keystone/keystone_service.h"
@@ -543,6 +547,96 @@ void bind_store(py::module_& m) {
         unwrap_void(e.recv(reinterpret_cast<void*>(ptr), n, peer, nullptr));
         unwrap_void(gpu::sync());
       });
+
+  // ------------------------------------------- collective batch shuffle
+  // wants: per peer (keys, sizes, recv_base device ptr)
+  m.def("gpu_batch_shuffle",
+        [](GpuClient& g, RcclEngine& e,
+           const std::vector<std::tuple<std::vector<std::string>,
+                                        std::vector<uint64_t>, uint64_t>>& wants) {
+          std::vector<ShuffleWant> w;
+          w.reserve(wants.size());
+          for (auto& [keys, sizes, base] : wants) {
+            ShuffleWant sw;
+            sw.keys = keys;
+            sw.sizes = sizes;
+            sw.recv_base = reinterpret_cast<void*>(base);
+            w.push_back(std::move(sw));
+          }
+          py::gil_scoped_release rel;
+          unwrap_void(g.batch_shuffle_rccl(e, w));
+        });
+
+  // Host-loopback shuffle (tests): N in-process "ranks" exchange through
+  // shared memory — exercises the full batch_shuffle algorithm (want-list
+  // encode/exchange, local resolution, gather layout, consistent-abort
+  // handshake, data all-to-all) with no GPU.
+  // objects[r]: key → bytes that rank r OWNS.
+  // want_keys[r][p]: keys rank r wants from rank p.
+  // Returns recv[r][p]: the concatenated payload rank r received from p.
+  m.def("loopback_shuffle_for_test",
+        [](const std::vector<std::map<std::string, py::bytes>>& objects,
+           const std::vector<std::vector<std::vector<std::string>>>& want_keys) {
+          const int n = static_cast<int>(objects.size());
+          // materialize owned objects as host buffers
+          std::vector<std::map<std::string, std::string>> own(n);
+          for (int r = 0; r < n; ++r)
+            for (auto& [k, v] : objects[r]) own[r][k] = std::string(v);
+          LoopbackGroup grp(n);
+          std::vector<std::vector<std::string>> recv(n);      // per rank, per peer
+          std::vector<std::vector<ShuffleWant>> wants(n);
+          for (int r = 0; r < n; ++r) {
+            recv[r].resize(n);
+            wants[r].resize(n);
+            for (int p = 0; p < n; ++p) {
+              auto& w = wants[r][p];
+              w.keys = want_keys[r][p];
+              uint64_t total = 0;
+              for (auto& k : w.keys) {
+                // sizes must be known to the requester (fixed-size pattern):
+                // look them up in the global object table
+                uint64_t sz = 0;
+                for (int q = 0; q < n; ++q) {
+                  auto it = own[q].find(k);
+                  if (it != own[q].end()) sz = it->second.size();
+                }
+                w.sizes.push_back(sz);
+                total += sz;
+              }
+              recv[r][p].resize(total);
+              w.recv_base = recv[r][p].data();
+            }
+          }
+          std::vector<std::string> errs(n);
+          {
+            py::gil_scoped_release rel;
+            std::vector<std::thread> ts;
+            for (int r = 0; r < n; ++r)
+              ts.emplace_back([&, r] {
+                auto ex = grp.exchanger(r);
+                HostCopier cp;
+                ShuffleResolver res = [&, r](const ObjectKey& k,
+                                             uint64_t sz) -> const void* {
+                  auto it = own[r].find(k);
+                  if (it == own[r].end() || it->second.size() != sz)
+                    return nullptr;
+                  return it->second.data();
+                };
+                auto rr = batch_shuffle(*ex, cp, res, wants[r]);
+                if (!rr.ok()) errs[r] = rr.message();
+              });
+            for (auto& t : ts) t.join();
+          }
+          for (int r = 0; r < n; ++r)
+            if (!errs[r].empty())
+              throw std::runtime_error("rank " + std::to_string(r) + ": " +
+                                       errs[r]);
+          std::vector<std::vector<py::bytes>> out(n);
+          for (int r = 0; r < n; ++r)
+            for (int p = 0; p < n; ++p)
+              out[r].emplace_back(recv[r][p]);
+          return out;
+        });
 
   m.def("client_batch_remove_prepared", [](Client& c, const DevPutBatch& b) {
     py::gil_scoped_release rel;

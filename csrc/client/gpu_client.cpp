@@ -1069,6 +1069,44 @@ Result<std::vector<int32_t>> GpuClient::batch_get_device(
   return statuses;
 }
 
+// --------------------------------------------------- collective shuffle
+
+Result<void> GpuClient::batch_shuffle_rccl(RcclEngine& comm,
+                                           const std::vector<ShuffleWant>& want) {
+  BB_RETURN_IF_ERROR(init());
+  BB_HIP(hipSetDevice(device_));
+  RcclExchanger ex(comm, streams_[3]);
+  GpuCopier cp(device_, streams_[4]);
+  // serve keys out of this rank's visible pools: placement cache first
+  // (covers this client's own puts), metadata RPC as the general fallback
+  ShuffleResolver resolve = [this](const ObjectKey& key,
+                                   uint64_t size) -> const void* {
+    bool cached = false;
+    CachedPlacement cp2;
+    {
+      std::lock_guard<std::mutex> g(cache_mu_);
+      auto it = placement_cache_.find(key);
+      if (it != placement_cache_.end() && it->second.size == size) {
+        cp2 = it->second;
+        cached = true;
+      }
+    }
+    if (cached) {  // device_pool_base may RPC — not under the lock
+      if (uint8_t* base = device_pool_base(cp2.pool_id)) return base + cp2.offset;
+      return nullptr;
+    }
+    auto meta = c_.meta_call<KeyMsg, GetWorkersResponse>(M::GET_WORKERS,
+                                                         KeyMsg{key});
+    if (!meta.ok() || meta->size != size) return nullptr;
+    for (const auto& copy : meta->copies) {
+      if (copy.shards.size() != 1) continue;
+      if (auto res = resolve_device_ptr(copy.shards[0]); res.ptr) return res.ptr;
+    }
+    return nullptr;
+  };
+  return batch_shuffle(ex, cp, resolve, want);
+}
+
 // ------------------------------------------------------- pipelined batches
 
 Result<uint64_t> GpuClient::batch_put_async(std::vector<DevPutItem> items,

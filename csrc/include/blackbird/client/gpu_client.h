@@ -19,9 +19,12 @@
 #include <unordered_map>
 
 #include "blackbird/client/client.h"
+#include "blackbird/client/shuffle.h"
 #include "blackbird/gpu/gpu_kernels.h"
 
 namespace blackbird {
+
+class RcclEngine;
 
 class GpuClient {
  public:
@@ -102,6 +105,14 @@ class GpuClient {
   // hipMemcpyAsync issue cost). set_fused_copy(false) falls back to SDMA
   // (hipMemcpyAsync on rotating streams) for every transfer.
   void set_fused_copy(bool on) { fused_copy_ = on; }
+
+  // ---- collective batch shuffle (RCCL over xGMI; see shuffle.h) ----
+  // Every rank calls this together: want[p] lists the objects this rank
+  // wants FROM rank p (received contiguously at want[p].recv_base). Served
+  // objects are resolved out of this rank's visible pools (placement cache
+  // first, metadata RPC fallback) and move in grouped all-to-all-v calls.
+  Result<void> batch_shuffle_rccl(RcclEngine& comm,
+                                  const std::vector<ShuffleWant>& want);
 
   // ---- verified placement cache (opt-in) ----
   // Remembers {pool, offset, size, digest} from this client's own puts and

@@ -715,3 +715,87 @@ class TestBatchSessions:
             c.close()
         finally:
             cl.stop()
+
+
+class TestRcclShuffle:
+    def test_two_rank_store_shuffle(self, tmp_path):
+        """batch_shuffle_rccl end to end: two ranks put objects into their
+        own HBM pools through the store, then exchange them with the RCCL
+        all-to-all shuffle (needs >=2 devices; the 8-GPU driver environment
+        runs it — the algorithm itself is covered CPU-side in
+        test_shuffle.py)."""
+        if bb.core.gpu.device_count() < 2:
+            pytest.skip("needs >=2 GPUs (RCCL: one rank per device)")
+        import subprocess, sys, textwrap
+        script = tmp_path / "shuf2.py"
+        script.write_text(textwrap.dedent(f"""
+            import sys, os, time
+            sys.path.insert(0, {repr(os.path.dirname(os.path.dirname(os.path.abspath(__file__))))})
+            import blackbird_amd as bb
+            rank = int(sys.argv[1]); coord_ep = sys.argv[2]; ks_ep = sys.argv[3]
+            g = bb.core.gpu
+            S, N = 65536, 32
+            coord = bb.CoordClient(); coord.connect(coord_ep)
+            wc = bb.WorkerConfig()
+            wc.worker_id = "w%d" % rank
+            wc.coord_endpoint = coord_ep
+            wc.data_listen_address = "127.0.0.1:0"
+            pc = bb.PoolConfig(); pc.pool_id = "hbm%d" % rank
+            pc.storage_class = bb.StorageClass.RAM_GPU
+            pc.size_bytes = 64 << 20; pc.gpu_device_id = rank
+            wc.pools = [pc]
+            w = bb.WorkerService(wc); w.initialize(); w.start()
+            o = bb.ClientOptions(); o.keystone_endpoint = ks_ep
+            c = bb.Client(o)
+            deadline = time.time() + 20
+            while True:
+                try:
+                    c.connect()
+                    if len(c.memory_pools()) >= 2: break
+                except Exception: pass
+                assert time.time() < deadline, "cluster did not assemble"
+                time.sleep(0.05)
+            gcl = bb.GpuClient(c, rank); gcl.init()
+            gcl.set_placement_cache(True)
+            cfg = bb.PlacementConfig()
+            cfg.preferred_worker = "w%d" % rank
+            src = g.malloc(N * S, rank)
+            g.fill_pattern(src, N * S, seed=500 + rank)
+            items = [("r%do%02d" % (rank, i), src + i * S, S) for i in range(N)]
+            assert gcl.batch_put_device(items, cfg) == [0] * N
+            # rendezvous then shuffle: each rank wants ALL peer objects
+            e = bb.RcclEngine()
+            e.init(coord, "shuftest", "t1", rank, 2, rank)
+            peer = 1 - rank
+            recv = g.malloc(N * S, rank)
+            wants = [([], [], 0), ([], [], 0)]
+            wants[peer] = (["r%do%02d" % (peer, i) for i in range(N)],
+                           [S] * N, recv)
+            bb.core.gpu_batch_shuffle(gcl, e, wants)
+            # received bytes are the peer's fill pattern
+            bad = g.verify_pattern(recv, N * S, seed=500 + peer)
+            assert bad == 0, bad
+            e.destroy()
+            c.close(); w.stop()
+            print("RANK", rank, "OK")
+        """))
+        cs = bb.CoordServer()
+        cs.start("127.0.0.1", 0)
+        ep = "127.0.0.1:%d" % cs.port
+        kc = bb.KeystoneConfig()
+        kc.listen_address = "127.0.0.1:0"
+        kc.coord_endpoint = ep
+        srv = bb.create_and_start_keystone(kc)
+        try:
+            procs = [subprocess.Popen(
+                         [sys.executable, str(script), str(r), ep, srv.endpoint],
+                         stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
+                     for r in (0, 1)]
+            for p in procs:
+                out, _ = p.communicate(timeout=180)
+                assert p.returncode == 0, out.decode()
+                assert b"OK" in out
+        finally:
+            srv.stop()
+            srv.service().stop()
+            cs.stop()
