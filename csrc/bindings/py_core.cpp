@@ -224,7 +224,9 @@ PYBIND11_MODULE(_core, m) {
       .def("load", [](coord::CoordStore& s, const std::string& path) {
         unwrap_void(s.load(path));
       })
-      .def("dirty", &coord::CoordStore::dirty);
+      .def("dirty", &coord::CoordStore::dirty)
+      .def("epoch", &coord::CoordStore::epoch)
+      .def("bump_epoch", &coord::CoordStore::bump_epoch);
 
   py::class_<coord::CoordService, std::shared_ptr<coord::CoordService>>(m, "CoordService")
       .def("put", [](coord::CoordService& c, const std::string& k,
@@ -327,6 +329,8 @@ PYBIND11_MODULE(_core, m) {
       .def("connect", [](coord::CoordClient& c, const std::string& ep) {
         unwrap_void(c.connect(ep));
       }, py::call_guard<py::gil_scoped_release>())
+      .def_property_readonly("observed_epoch",
+                             &coord::CoordClient::observed_epoch)
       .def("close", &coord::CoordClient::close,
            py::call_guard<py::gil_scoped_release>());
 

@@ -61,7 +61,8 @@ Result<void> CoordStore::save(const std::string& path) {
   serde::Enc e;
   {
     std::lock_guard<std::mutex> g(mu_);
-    e.num<uint32_t>(0xBBC05EED);  // magic + version
+    e.num<uint32_t>(0xBBC05EEE);  // magic + version (v2: carries epoch)
+    e.num<uint64_t>(epoch_.load());
     e.num<uint32_t>(static_cast<uint32_t>(kv_.size()));
     for (const auto& [k, entry] : kv_) {
       e.str(k);
@@ -93,8 +94,12 @@ Result<void> CoordStore::load(const std::string& path) {
   while ((n = fread(chunk, 1, sizeof(chunk), f)) > 0) buf.append(chunk, n);
   fclose(f);
   serde::Dec d(buf.data(), buf.size());
-  if (d.num<uint32_t>() != 0xBBC05EED)
+  const uint32_t magic = d.num<uint32_t>();
+  if (magic == 0xBBC05EEE) {
+    epoch_.store(d.num<uint64_t>());
+  } else if (magic != 0xBBC05EED) {  // v1: no epoch field
     return Error{ErrorCode::PROTOCOL_ERROR, "bad snapshot magic in " + path};
+  }
   uint32_t count = d.num<uint32_t>();
   const uint64_t now = now_ms();
   std::lock_guard<std::mutex> g(mu_);
@@ -290,17 +295,50 @@ Result<Req> decode(const std::string& body) {
     return Error{ErrorCode::PROTOCOL_ERROR, "bad request body"};
   return r;
 }
+
+// mutation requests carry a trailing fencing epoch (u64; absent = 0,
+// accepted for fencing-unaware callers — embedded/in-process paths)
+template <typename Req>
+Result<std::pair<Req, uint64_t>> decode_fenced(const std::string& body) {
+  serde::Dec d(body.data(), body.size());
+  Req r{};
+  serde::get(d, r);
+  if (!d.ok()) return Error{ErrorCode::PROTOCOL_ERROR, "bad request body"};
+  uint64_t ep = d.remaining() >= 8 ? d.num<uint64_t>() : 0;
+  return std::make_pair(std::move(r), ep);
+}
 }  // namespace
+
+Result<void> CoordServer::check_writable(uint64_t client_epoch) {
+  const uint64_t mine = store_->epoch();
+  if (client_epoch > mine) {
+    // a newer leadership generation exists: this server was deposed while it
+    // was down/partitioned. Fence permanently — from now on EVERY client
+    // (stale ones included) gets NOT_LEADER here and rediscovers.
+    if (!read_only_.exchange(true))
+      BB_LOG(WARN) << "coordd fenced: local epoch " << mine
+                   << " superseded by client-carried epoch " << client_epoch;
+    return Error{ErrorCode::NOT_LEADER,
+                 "fenced: superseded leadership (epoch=" +
+                     std::to_string(client_epoch) + ")"};
+  }
+  if (read_only_.load())
+    return Error{ErrorCode::NOT_LEADER,
+                 "standby coordd (read-only) (epoch=" + std::to_string(mine) + ")"};
+  // client_epoch < mine is fine: a client that has not yet observed the new
+  // leadership is writing to the CURRENT leader — fencing only exists to
+  // refuse the DEPOSED one (client_epoch > mine above)
+  return {};
+}
 
 CoordServer::CoordServer(std::shared_ptr<CoordStore> store) : store_(std::move(store)) {
   using Ctx = rpc::RpcServer::ConnCtx;
 
   rpc_.register_handler(method::PUT, [this](const std::string& b, const Ctx&) -> Result<std::string> {
-    if (read_only_.load())
-      return Error{ErrorCode::NOT_LEADER, "standby coordd (read-only)"};
-    auto r = decode<PutReq>(b);
+    auto r = decode_fenced<PutReq>(b);
     if (!r.ok()) return r.error();
-    BB_RETURN_IF_ERROR(store_->put(r->key, r->value, r->ttl_ms));
+    BB_RETURN_IF_ERROR(check_writable(r->second));
+    BB_RETURN_IF_ERROR(store_->put(r->first.key, r->first.value, r->first.ttl_ms));
     return std::string{};
   });
   rpc_.register_handler(method::GET, [this](const std::string& b, const Ctx&) -> Result<std::string> {
@@ -311,11 +349,10 @@ CoordServer::CoordServer(std::shared_ptr<CoordStore> store) : store_(std::move(s
     return serde::to_bytes(ValueResp{v.value()});
   });
   rpc_.register_handler(method::DEL, [this](const std::string& b, const Ctx&) -> Result<std::string> {
-    if (read_only_.load())
-      return Error{ErrorCode::NOT_LEADER, "standby coordd (read-only)"};
-    auto r = decode<KeyReq>(b);
+    auto r = decode_fenced<KeyReq>(b);
     if (!r.ok()) return r.error();
-    BB_RETURN_IF_ERROR(store_->del(r->key));
+    BB_RETURN_IF_ERROR(check_writable(r->second));
+    BB_RETURN_IF_ERROR(store_->del(r->first.key));
     return std::string{};
   });
   rpc_.register_handler(method::GET_PREFIX, [this](const std::string& b, const Ctx&) -> Result<std::string> {
@@ -326,20 +363,19 @@ CoordServer::CoordServer(std::shared_ptr<CoordStore> store) : store_(std::move(s
     return serde::to_bytes(PrefixResp{std::move(v.value())});
   });
   rpc_.register_handler(method::CAS, [this](const std::string& b, const Ctx&) -> Result<std::string> {
-    if (read_only_.load())
-      return Error{ErrorCode::NOT_LEADER, "standby coordd (read-only)"};
-    auto r = decode<CasReq>(b);
+    auto r = decode_fenced<CasReq>(b);
     if (!r.ok()) return r.error();
-    auto v = store_->cas(r->key, r->expected, r->expect_absent != 0, r->value, r->ttl_ms);
+    BB_RETURN_IF_ERROR(check_writable(r->second));
+    auto& rq = r->first;
+    auto v = store_->cas(rq.key, rq.expected, rq.expect_absent != 0, rq.value, rq.ttl_ms);
     if (!v.ok()) return v.error();
     return serde::to_bytes(BoolResp{static_cast<uint8_t>(v.value() ? 1 : 0)});
   });
   rpc_.register_handler(method::KEEPALIVE, [this](const std::string& b, const Ctx&) -> Result<std::string> {
-    if (read_only_.load())
-      return Error{ErrorCode::NOT_LEADER, "standby coordd (read-only)"};
-    auto r = decode<KeepAliveReq>(b);
+    auto r = decode_fenced<KeepAliveReq>(b);
     if (!r.ok()) return r.error();
-    BB_RETURN_IF_ERROR(store_->keep_alive(r->key, r->ttl_ms));
+    BB_RETURN_IF_ERROR(check_writable(r->second));
+    BB_RETURN_IF_ERROR(store_->keep_alive(r->first.key, r->first.ttl_ms));
     return std::string{};
   });
   rpc_.register_handler(method::WATCH, [this](const std::string& b, const Ctx& ctx) -> Result<std::string> {
@@ -368,7 +404,14 @@ CoordServer::CoordServer(std::shared_ptr<CoordStore> store) : store_(std::move(s
 
   rpc_.register_handler(method::DUMP, [this](const std::string&, const Ctx&) -> Result<std::string> {
     serde::Enc e;
+    e.num<uint64_t>(store_->epoch());
     serde::put(e, store_->dump());
+    return std::move(e.buf);
+  });
+
+  rpc_.register_handler(method::EPOCH, [this](const std::string&, const Ctx&) -> Result<std::string> {
+    serde::Enc e;
+    e.num<uint64_t>(store_->epoch());
     return std::move(e.buf);
   });
 
@@ -420,6 +463,28 @@ void CoordClient::install_event_callback() {
   });
 }
 
+std::string CoordClient::fenced(std::string body) {
+  uint64_t e = epoch_.load();
+  body.append(reinterpret_cast<const char*>(&e), sizeof(e));
+  return body;
+}
+
+void CoordClient::observe_epoch(uint64_t e) {
+  uint64_t cur = epoch_.load();
+  while (e > cur && !epoch_.compare_exchange_weak(cur, e)) {
+  }
+}
+
+// best-effort fetch of the server's fencing epoch (callers hold mu_ or are
+// in connect()); never lowers the observed epoch
+void CoordClient::refresh_epoch_locked() {
+  auto r = rpc_.call_raw(method::EPOCH, {});
+  if (!r.ok() || r.value().size() < 8) return;
+  uint64_t e = 0;
+  std::memcpy(&e, r.value().data(), 8);
+  observe_epoch(e);
+}
+
 Result<void> CoordClient::connect(const std::string& endpoint, int timeout_ms) {
   std::vector<std::string> eps;
   size_t start = 0;
@@ -442,6 +507,7 @@ Result<void> CoordClient::connect(const std::string& endpoint, int timeout_ms) {
     auto r = rpc_.connect(endpoints_[ep_cursor_], timeout_ms);
     if (r.ok()) {
       install_event_callback();
+      refresh_epoch_locked();
       return {};
     }
     last = r.error();
@@ -467,6 +533,7 @@ Result<void> CoordClient::redial_locked() {
   }
   if (!ok) return last;
   install_event_callback();
+  refresh_epoch_locked();
   // re-subscribe every watch on the fresh connection
   server_to_client_.clear();
   for (auto& [cid, entry] : watches_) {
@@ -495,6 +562,13 @@ Result<std::string> CoordClient::call_with_retry(uint16_t m, const std::string& 
   }
   for (size_t attempt = 0; attempt < hops; ++attempt) {
     bool standby = r.code() == ErrorCode::NOT_LEADER;
+    if (standby) {
+      // NOT_LEADER messages carry the server's view of the fencing epoch as
+      // "(epoch=N)" — track the max so a revived stale primary gets fenced
+      auto pos = r.message().find("epoch=");
+      if (pos != std::string::npos)
+        observe_epoch(strtoull(r.message().c_str() + pos + 6, nullptr, 10));
+    }
     switch (r.code()) {
       case ErrorCode::NOT_CONNECTED:
       case ErrorCode::CONNECTION_CLOSED:
@@ -524,7 +598,7 @@ Result<std::string> CoordClient::call_with_retry(uint16_t m, const std::string& 
 }
 
 Result<void> CoordClient::put(const std::string& k, const std::string& v, uint64_t ttl) {
-  auto r = call_with_retry(method::PUT, serde::to_bytes(PutReq{k, v, ttl}));
+  auto r = call_with_retry(method::PUT, fenced(serde::to_bytes(PutReq{k, v, ttl})));
   if (!r.ok()) return r.error();
   return {};
 }
@@ -539,7 +613,7 @@ Result<std::string> CoordClient::get(const std::string& k) {
 }
 
 Result<void> CoordClient::del(const std::string& k) {
-  auto r = call_with_retry(method::DEL, serde::to_bytes(KeyReq{k}));
+  auto r = call_with_retry(method::DEL, fenced(serde::to_bytes(KeyReq{k})));
   if (!r.ok()) return r.error();
   return {};
 }
@@ -556,7 +630,8 @@ Result<std::vector<KV>> CoordClient::get_prefix(const std::string& p) {
 Result<bool> CoordClient::cas(const std::string& k, const std::string& e, bool ea,
                               const std::string& v, uint64_t ttl) {
   auto r = call_with_retry(
-      method::CAS, serde::to_bytes(CasReq{k, e, v, static_cast<uint8_t>(ea), ttl}));
+      method::CAS,
+      fenced(serde::to_bytes(CasReq{k, e, v, static_cast<uint8_t>(ea), ttl})));
   if (!r.ok()) return r.error();
   BoolResp resp;
   if (!serde::from_bytes(r.value(), resp))
@@ -565,7 +640,8 @@ Result<bool> CoordClient::cas(const std::string& k, const std::string& e, bool e
 }
 
 Result<void> CoordClient::keep_alive(const std::string& k, uint64_t ttl) {
-  auto r = call_with_retry(method::KEEPALIVE, serde::to_bytes(KeepAliveReq{k, ttl}));
+  auto r = call_with_retry(method::KEEPALIVE,
+                           fenced(serde::to_bytes(KeepAliveReq{k, ttl})));
   if (!r.ok()) return r.error();
   return {};
 }
@@ -701,9 +777,19 @@ Result<void> CoordFollower::sync_once() {
   auto d = rpc_.call_raw(method::DUMP, {});
   if (!d.ok()) return d.error();
   serde::Dec dec(d.value().data(), d.value().size());
+  const uint64_t remote_epoch = dec.num<uint64_t>();
   std::vector<DumpEntry> entries;
   serde::get(dec, entries);
   if (!dec.ok()) return Error{ErrorCode::PROTOCOL_ERROR, "bad DUMP"};
+  // never re-follow a DEPOSED primary: if we promoted (or synced from a
+  // newer leadership) our epoch is higher — mirroring an older generation
+  // would resurrect the split brain the fencing epoch exists to prevent
+  if (remote_epoch < store_->epoch())
+    return Error{ErrorCode::NOT_LEADER,
+                 "primary " + primary_ + " carries stale epoch " +
+                     std::to_string(remote_epoch) + " < " +
+                     std::to_string(store_->epoch())};
+  store_->set_epoch(remote_epoch);
   for (auto& e : entries) (void)store_->put(e.key, e.value, e.ttl_ms);
   BB_LOG(INFO) << "coordd follower synced " << entries.size()
                << " keys from " << primary_;
@@ -749,8 +835,13 @@ void CoordFollower::monitor_loop() {
       continue;
     }
     if (now - down_since >= failover_ms_) {
+      // bump the fencing epoch FIRST: every mutation accepted from here on
+      // carries the new generation, so a revived old primary self-fences
+      // the moment any failed-over client touches it
+      const uint64_t e = store_->bump_epoch();
       BB_LOG(WARN) << "coordd follower PROMOTING after "
-                   << (now - down_since) << " ms without a primary";
+                   << (now - down_since) << " ms without a primary (epoch "
+                   << e << ")";
       promoted_ = true;
       if (server_) server_->set_read_only(false);
       running_ = false;

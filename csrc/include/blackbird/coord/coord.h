@@ -100,6 +100,23 @@ class CoordStore {
   Result<void> load(const std::string& path);
   // True if a put/del/cas/keep_alive happened since the last save().
   bool dirty() const { return dirty_.load(); }
+
+  // ---- leadership fencing epoch ----
+  // Monotonic leadership generation. A follower bumps it when it PROMOTES;
+  // clients carry it on every mutation; a server that sees a NEWER epoch
+  // than its own has been superseded and permanently demotes itself to
+  // read-only (the revived-old-primary split-brain window of async
+  // replication — etcd got this from raft terms; this is the fencing-token
+  // equivalent). Persisted in snapshots, transferred in DUMP.
+  uint64_t epoch() const { return epoch_.load(); }
+  void set_epoch(uint64_t e) {
+    epoch_.store(e);
+    dirty_ = true;
+  }
+  uint64_t bump_epoch() {
+    dirty_ = true;
+    return epoch_.fetch_add(1) + 1;
+  }
   // Full state with remaining TTLs (follower bootstrap; expired entries
   // skipped).
   std::vector<DumpEntry> dump();
@@ -121,6 +138,7 @@ class CoordStore {
   std::mutex mu_;
   std::map<std::string, Entry> kv_;
   std::map<uint64_t, Watch> watches_;
+  std::atomic<uint64_t> epoch_{1};
   std::atomic<bool> dirty_{false};
   std::atomic<uint64_t> next_watch_{1};
   std::atomic<bool> running_{true};
@@ -172,7 +190,8 @@ constexpr uint16_t CAS = 104;
 constexpr uint16_t KEEPALIVE = 105;
 constexpr uint16_t WATCH = 106;
 constexpr uint16_t UNWATCH = 107;
-constexpr uint16_t DUMP = 108;  // full state incl. TTLs (replication)
+constexpr uint16_t DUMP = 108;  // full state incl. TTLs + epoch (replication)
+constexpr uint16_t EPOCH = 109;  // current fencing epoch (u64)
 }  // namespace method
 
 class CoordServer {
@@ -190,6 +209,9 @@ class CoordServer {
   bool read_only() const { return read_only_.load(); }
 
  private:
+  // write-barrier: read_only + fencing epoch (see CoordStore::epoch)
+  Result<void> check_writable(uint64_t client_epoch);
+
   std::shared_ptr<CoordStore> store_;
   rpc::RpcServer rpc_;
   std::atomic<bool> read_only_{false};
@@ -257,10 +279,17 @@ class CoordClient : public CoordService {
   Result<void> keep_alive(const std::string& k, uint64_t ttl) override;
   Result<uint64_t> watch_prefix(const std::string& p, WatchCallback cb) override;
   Result<void> unwatch(uint64_t id) override;
+  // highest leadership epoch observed (sent with every mutation; a stale
+  // revived primary rejects + self-fences on seeing a newer one)
+  uint64_t observed_epoch() const { return epoch_.load(); }
 
  private:
   // retry wrapper: redial + re-subscribe watches on connection loss
   Result<std::string> call_with_retry(uint16_t method, const std::string& body);
+  // mutation body + trailing fencing epoch
+  std::string fenced(std::string body);
+  void refresh_epoch_locked();
+  void observe_epoch(uint64_t e);
   Result<void> redial_locked();
   void install_event_callback();
 
@@ -278,6 +307,7 @@ class CoordClient : public CoordService {
   std::map<uint64_t, uint64_t> server_to_client_;  // server id → client id
   uint64_t next_client_watch_ = 1;
   std::function<void()> on_reconnect_;
+  std::atomic<uint64_t> epoch_{0};  // max leadership epoch seen
 };
 
 // Build the right client for an endpoint ("" = fresh embedded store).

@@ -495,3 +495,96 @@ class TestReplication:
             srv.stop()
             srv.service().stop()
             b.stop()
+
+
+class TestFencing:
+    """Leadership fencing epochs: a deposed primary that
+    revives AFTER standby promotion must not accept writes. Clients carry the
+    highest epoch they observed; a server seeing a newer epoch than its own
+    permanently demotes itself."""
+
+    def test_revived_primary_is_fenced(self):
+        a = bb.CoordServer()
+        a.start("127.0.0.1", 0)
+        ep_a = "127.0.0.1:%d" % a.port
+        store_a = a.store()
+        b = bb.CoordServer()
+        b.start("127.0.0.1", 0)
+        ep_b = "127.0.0.1:%d" % b.port
+        a2 = None
+        c = bb.CoordClient()
+        c2 = bb.CoordClient()
+        try:
+            f = bb.CoordFollower(b, ep_a, failover_ms=400)
+            f.start()
+            c.connect("%s,%s" % (ep_a, ep_b))
+            epoch0 = store_a.epoch()
+            c.put("/z/k", "v1")
+            time.sleep(0.3)  # replication streams it
+            a.stop()
+            deadline = time.time() + 6
+            while time.time() < deadline and not f.promoted():
+                time.sleep(0.05)
+            assert f.promoted()
+            # promotion bumped the fencing epoch
+            assert b.store().epoch() == epoch0 + 1
+            c.put("/z/k", "v2")  # client fails over to B, observes new epoch
+            assert c.observed_epoch == epoch0 + 1
+
+            # REVIVE the old primary: same store (stale epoch), new listener
+            a2 = bb.CoordServer(store_a)
+            a2.start("127.0.0.1", 0)
+            ep_a2 = "127.0.0.1:%d" % a2.port
+            # the failed-over client reaches the revived primary: its write
+            # carries the newer epoch and must be REFUSED
+            c.connect(ep_a2)
+            with pytest.raises(Exception, match="NOT_LEADER"):
+                c.put("/z/k", "v3-split-brain")
+            # ...and the refusal FENCED the revived primary permanently:
+            # even a fresh, epoch-ignorant client is refused now
+            assert a2.read_only()
+            c2.connect(ep_a2)
+            with pytest.raises(Exception, match="NOT_LEADER"):
+                c2.put("/z/other", "x")
+            # the split-brain write never landed anywhere
+            assert b.store().get("/z/k") == "v2"
+            with pytest.raises(Exception):
+                store_a.get("/z/other")
+            f.stop()
+        finally:
+            c.close()
+            c2.close()
+            if a2 is not None:
+                a2.stop()
+            b.stop()
+            a.stop()
+
+    def test_follower_refuses_stale_primary(self):
+        # a promoted store (higher epoch) must never re-follow an old-epoch
+        # primary — mirroring it would resurrect the split brain
+        a = bb.CoordServer()
+        a.start("127.0.0.1", 0)
+        ep_a = "127.0.0.1:%d" % a.port
+        b = bb.CoordServer()
+        b.start("127.0.0.1", 0)
+        try:
+            b.store().bump_epoch()  # b was promoted at some point
+            f = bb.CoordFollower(b, ep_a, failover_ms=60000)
+            with pytest.raises(Exception, match="stale epoch|NOT_LEADER"):
+                f.start()
+        finally:
+            b.stop()
+            a.stop()
+
+    def test_epoch_persists_in_snapshot(self, tmp_path):
+        s = bb.CoordStore()
+        s.put("/p/k", "v")
+        s.bump_epoch()
+        s.bump_epoch()
+        e = s.epoch()
+        path = str(tmp_path / "snap.bb")
+        s.save(path)
+        s2 = bb.CoordStore()
+        s2.load(path)
+        assert s2.epoch() == e
+        assert s2.get("/p/k") == "v"
