@@ -97,7 +97,8 @@ PYBIND11_MODULE(_core, m) {
       .def_readwrite("storage_class", &ShardPlacement::storage_class)
       .def_readwrite("offset", &ShardPlacement::offset)
       .def_readwrite("length", &ShardPlacement::length)
-      .def_readwrite("access", &ShardPlacement::access);
+      .def_readwrite("access", &ShardPlacement::access)
+      .def_readwrite("digest", &ShardPlacement::digest);
 
   py::class_<CopyPlacement>(m, "CopyPlacement")
       .def(py::init<>())

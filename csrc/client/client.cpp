@@ -395,6 +395,32 @@ Result<void> Client::read_copy(const std::vector<CopyPlacement>& copies,
 
 // ------------------------------------------------------------ object ops
 
+namespace {
+// Per-copy per-shard standalone digests for striped copies (empty when every
+// copy is single-shard — the whole-object checksum covers those). Recorded
+// at put time so the scrubber can verify each striped shard independently.
+std::vector<std::vector<uint64_t>> host_shard_digests(
+    const std::vector<CopyPlacement>& copies, const void* data) {
+  bool any_striped = false;
+  for (const auto& c : copies)
+    if (c.shards.size() > 1) any_striped = true;
+  if (!any_striped) return {};
+  std::vector<std::vector<uint64_t>> out;
+  out.reserve(copies.size());
+  for (const auto& c : copies) {
+    std::vector<uint64_t> ds;
+    uint64_t off = 0;
+    for (const auto& s : c.shards) {
+      ds.push_back(gpu::checksum_cpu(
+          static_cast<const uint8_t*>(data) + off, s.length));
+      off += s.length;
+    }
+    out.push_back(std::move(ds));
+  }
+  return out;
+}
+}  // namespace
+
 Result<void> Client::put(const ObjectKey& key, const void* data, uint64_t size,
                          const PlacementConfig& cfg) {
   PutStartRequest req{key, size, cfg};
@@ -408,8 +434,10 @@ Result<void> Client::put(const ObjectKey& key, const void* data, uint64_t size,
     return xfer.error();
   }
   uint64_t checksum = cfg.checksum ? gpu::checksum_cpu(data, size) : 0;
-  auto done = meta_call_raw(M::PUT_COMPLETE,
-                             serde::to_bytes(PutCompleteRequest{key, checksum}),
+  PutCompleteRequest done_req{key, checksum};
+  if (cfg.checksum)
+    done_req.shard_digests = host_shard_digests(start->copies, data);
+  auto done = meta_call_raw(M::PUT_COMPLETE, serde::to_bytes(done_req),
                              opts_.rpc_timeout_ms);
   if (!done.ok()) return done.error();
   return {};
@@ -539,7 +567,11 @@ Result<std::vector<int32_t>> Client::batch_put(const std::vector<PutItem>& items
           cancels2.push_back(items[i].key);
           continue;
         }
-        completes2.reqs.push_back(PutCompleteRequest{items[i].key, digests[i]});
+        PutCompleteRequest pc{items[i].key, digests[i]};
+        if (cfg.checksum)
+          pc.shard_digests =
+              host_shard_digests(start->items[i].copies, items[i].data);
+        completes2.reqs.push_back(std::move(pc));
       }
       if (!completes2.reqs.empty()) {
         auto r = meta_call<PutCompleteListMsg, StatusListMsg>(
@@ -587,7 +619,11 @@ Result<std::vector<int32_t>> Client::batch_put(const std::vector<PutItem>& items
       cancels.push_back(items[i].key);
       continue;
     }
-    completes.reqs.push_back(PutCompleteRequest{items[i].key, digests[i]});
+    PutCompleteRequest pc{items[i].key, digests[i]};
+    if (cfg.checksum)
+      pc.shard_digests =
+          host_shard_digests(start->items[i].copies, items[i].data);
+    completes.reqs.push_back(std::move(pc));
   }
   if (!completes.reqs.empty()) {
     auto r = meta_call<PutCompleteListMsg, StatusListMsg>(

@@ -4,6 +4,7 @@
 #include <atomic>
 #include <cstring>
 #include <future>
+#include <map>
 
 #include <hip/hip_runtime_api.h>
 
@@ -948,10 +949,53 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device(
   }
   for (auto& st : streams_) BB_HIP(hipStreamSynchronize(st));
 
+  // per-shard digests for striped copies (one batched launch hashes every
+  // source slice) — recorded server-side so the scrubber can verify each
+  // striped shard independently
+  std::map<uint32_t, std::vector<std::vector<uint64_t>>> shard_digests;
+  if (cfg.checksum) {
+    std::vector<const void*> sptrs;
+    std::vector<uint64_t> ssizes;
+    std::vector<std::tuple<uint32_t, uint32_t, uint32_t>> slots;  // item,copy,shard
+    for (auto i : committed_idx) {
+      const auto& copies = start->items[i].copies;
+      bool striped = false;
+      for (const auto& c : copies)
+        if (c.shards.size() > 1) striped = true;
+      if (!striped) continue;
+      auto& out = shard_digests[i];
+      out.resize(copies.size());
+      for (uint32_t c = 0; c < copies.size(); ++c) {
+        out[c].resize(copies[c].shards.size(), 0);
+        uint64_t off = 0;
+        for (uint32_t s2 = 0; s2 < copies[c].shards.size(); ++s2) {
+          sptrs.push_back(static_cast<const uint8_t*>(items[i].ptr) + off);
+          ssizes.push_back(copies[c].shards[s2].length);
+          slots.emplace_back(i, c, s2);
+          off += copies[c].shards[s2].length;
+        }
+      }
+    }
+    if (!sptrs.empty()) {
+      std::vector<uint64_t> got(sptrs.size(), 0);
+      auto r = gpu::checksum_batch(sptrs.data(), ssizes.data(),
+                                   static_cast<uint32_t>(sptrs.size()),
+                                   got.data(), device_, streams_[1]);
+      if (!r.ok()) return r.error();
+      for (size_t k = 0; k < slots.size(); ++k) {
+        auto [i, c, s2] = slots[k];
+        shard_digests[i][c][s2] = got[k];
+      }
+    }
+  }
+
   PutCompleteListMsg completes;
-  for (size_t j = 0; j < committed_idx.size(); ++j)
-    completes.reqs.push_back(
-        PutCompleteRequest{items[committed_idx[j]].key, digests[j]});
+  for (size_t j = 0; j < committed_idx.size(); ++j) {
+    PutCompleteRequest pc{items[committed_idx[j]].key, digests[j]};
+    auto it = shard_digests.find(committed_idx[j]);
+    if (it != shard_digests.end()) pc.shard_digests = std::move(it->second);
+    completes.reqs.push_back(std::move(pc));
+  }
   for (size_t j = 0; j < fused_hash_idx.size(); ++j)
     completes.reqs.push_back(
         PutCompleteRequest{items[fused_hash_idx[j]].key, fused_digests[j]});

@@ -177,8 +177,13 @@ struct ShardPlacement {
   uint64_t offset = 0;  // pool-relative byte offset
   uint64_t length = 0;
   AccessInfo access;
+  // standalone bbhash64 of THIS shard's bytes (0 = not recorded). Recorded
+  // at put time so the scrubber can verify striped copies shard by shard —
+  // the whole-object digest is position-weighted and not combinable across
+  // shard boundaries.
+  uint64_t digest = 0;
 
-  BB_FIELDS(pool_id, worker_id, storage_class, offset, length, access)
+  BB_FIELDS(pool_id, worker_id, storage_class, offset, length, access, digest)
 
   bool operator==(const ShardPlacement& o) const {
     return pool_id == o.pool_id && offset == o.offset && length == o.length;
@@ -380,7 +385,10 @@ struct PutStartResponse {
 struct PutCompleteRequest {
   ObjectKey key;
   uint64_t checksum = 0;
-  BB_FIELDS(key, checksum)
+  // per copy, per shard: standalone digests of the striped pieces (empty
+  // for single-shard puts — the whole-object checksum covers those)
+  std::vector<std::vector<uint64_t>> shard_digests;
+  BB_FIELDS(key, checksum, shard_digests)
 };
 
 struct GetWorkersResponse {

@@ -46,7 +46,9 @@ class KeystoneService {
   Result<GetWorkersResponse> get_workers(const ObjectKey& key);
   Result<PutStartResponse> put_start(const ObjectKey& key, uint64_t size,
                                      const PlacementConfig& cfg);
-  Result<void> put_complete(const ObjectKey& key, uint64_t checksum);
+  Result<void> put_complete(
+      const ObjectKey& key, uint64_t checksum,
+      const std::vector<std::vector<uint64_t>>& shard_digests = {});
   Result<void> put_cancel(const ObjectKey& key);
   Result<void> remove_object(const ObjectKey& key);
   uint64_t remove_all_objects();

@@ -94,7 +94,7 @@ void KeystoneServer::register_handlers() {
   rpc_.register_handler(M::PUT_COMPLETE, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
     auto r = decode<PutCompleteRequest>(b);
     if (!r.ok()) return r.error();
-    BB_RETURN_IF_ERROR(ks.put_complete(r->key, r->checksum));
+    BB_RETURN_IF_ERROR(ks.put_complete(r->key, r->checksum, r->shard_digests));
     return std::string{};
   });
   rpc_.register_handler(M::PUT_CANCEL, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {

@@ -166,7 +166,25 @@ Result<PutStartResponse> KeystoneService::put_start(const ObjectKey& key,
   return resp;
 }
 
-Result<void> KeystoneService::put_complete(const ObjectKey& key, uint64_t checksum) {
+namespace {
+// Record per-shard digests into the copies; single-shard copies default to
+// the whole-object checksum (their shard IS the object).
+void apply_shard_digests(ObjectMeta& m, uint64_t checksum,
+                         const std::vector<std::vector<uint64_t>>& sd) {
+  for (size_t c = 0; c < m.copies.size(); ++c) {
+    auto& shards = m.copies[c].shards;
+    if (c < sd.size() && sd[c].size() == shards.size()) {
+      for (size_t s = 0; s < shards.size(); ++s) shards[s].digest = sd[c][s];
+    } else if (shards.size() == 1) {
+      shards[0].digest = checksum;
+    }
+  }
+}
+}  // namespace
+
+Result<void> KeystoneService::put_complete(
+    const ObjectKey& key, uint64_t checksum,
+    const std::vector<std::vector<uint64_t>>& shard_digests) {
   std::unique_lock lk(objects_mu_);
   auto it = objects_.find(key);
   if (it == objects_.end()) return Error{ErrorCode::OBJECT_NOT_FOUND, key};
@@ -174,6 +192,7 @@ Result<void> KeystoneService::put_complete(const ObjectKey& key, uint64_t checks
     return Error{ErrorCode::INVALID_STATE, "already committed: " + key};
   it->second.state = ObjectState::COMMITTED;
   it->second.checksum = checksum;
+  apply_shard_digests(it->second, checksum, shard_digests);
   it->second.created_ms = now_ms();  // TTL starts at commit
   it->second.last_access_ms = it->second.created_ms;
   mark_dirty_locked(key, false);
@@ -393,6 +412,7 @@ std::vector<int32_t> KeystoneService::batch_put_complete(
       }
       it->second.state = ObjectState::COMMITTED;
       it->second.checksum = reqs[i].checksum;
+      apply_shard_digests(it->second, reqs[i].checksum, reqs[i].shard_digests);
       it->second.created_ms = now;
       it->second.last_access_ms = now;
       mark_dirty_locked(reqs[i].key, false);
@@ -495,6 +515,8 @@ Result<void> KeystoneService::commit_token(
     ObjectMeta* m = s->metas[i];
     m->state = ObjectState::COMMITTED;
     m->checksum = digests[i];
+    // sessions cover single-copy single-shard objects by construction
+    m->copies[0].shards[0].digest = digests[i];
     m->created_ms = now;  // TTL restarts at commit, as in put_complete
     m->last_access_ms = now;
     if (persist) mark_dirty_locked(m->key, false);
@@ -858,6 +880,9 @@ Result<void> KeystoneService::migrate_object(const ObjectKey& key,
       BB_LOG(ERROR) << "migration rename failed: " << rn.message();
     }
     it->second.copies = std::move(placed.value());
+    if (it->second.copies.size() == 1 &&
+        it->second.copies[0].shards.size() == 1)
+      it->second.copies[0].shards[0].digest = it->second.checksum;
     it->second.access_count = 0;
     bump_placement_epoch_locked();
     mark_dirty_locked(key, false);
@@ -1242,7 +1267,11 @@ uint32_t KeystoneService::run_scrub_once(uint32_t max_objects) {
   struct Cand {
     ObjectKey key;
     uint64_t checksum;
-    std::vector<std::vector<ShardPlacement>> copies;  // single-shard each
+    // every copy, striped or not: shard digests recorded at put time make
+    // each shard independently verifiable (the round-1 scrubber silently
+    // skipped striped copies — exactly the objects with the most failure
+    // surface)
+    std::vector<std::vector<ShardPlacement>> copies;
   };
   std::vector<Cand> cands;
   {
@@ -1254,10 +1283,7 @@ uint32_t KeystoneService::run_scrub_once(uint32_t max_objects) {
       if (meta.last_scrub_ms != 0 &&
           now < meta.last_scrub_ms + config_.scrub_interval_ms)
         continue;
-      bool all_single = !meta.copies.empty();
-      for (const auto& c : meta.copies)
-        if (c.shards.size() != 1) all_single = false;
-      if (!all_single) continue;
+      if (meta.copies.empty()) continue;
       meta.last_scrub_ms = now;
       Cand cd;
       cd.key = key;
@@ -1270,23 +1296,41 @@ uint32_t KeystoneService::run_scrub_once(uint32_t max_objects) {
   uint32_t quarantined = 0;
   for (const auto& cd : cands) {
     for (const auto& shards : cd.copies) {
-      const auto& sh = shards[0];
-      auto access = allocator_.pool_access(sh.pool_id);
-      if (!access.ok()) continue;  // pool gone: dead-worker cleanup owns it
-      auto* dc = data_client(access.value().endpoint);
-      if (!dc) continue;
-      ScrubChecksumReq req{sh.pool_id, sh.offset, sh.length};
-      auto resp = dc->call_raw(rpc::methods::DATA_CHECKSUM,
-                               serde::to_bytes(req), 60000);
-      if (!resp.ok()) continue;  // transient worker trouble: retry next pass
-      ScrubU64Msg got;
-      if (!serde::from_bytes(resp.value(), got)) continue;
-      if (got.v == cd.checksum) continue;
+      // verify every shard of the copy; ONE bad shard quarantines the copy
+      bool copy_bad = false;
+      bool verified_any = false;
+      const ShardPlacement* bad_sh = nullptr;
+      uint64_t bad_want = 0, bad_got = 0;
+      for (const auto& sh : shards) {
+        const uint64_t want =
+            sh.digest != 0 ? sh.digest
+                           : (shards.size() == 1 ? cd.checksum : 0);
+        if (want == 0) continue;  // digest never recorded (e.g. repair copy)
+        auto access = allocator_.pool_access(sh.pool_id);
+        if (!access.ok()) continue;  // pool gone: dead-worker cleanup owns it
+        auto* dc = data_client(access.value().endpoint);
+        if (!dc) continue;
+        ScrubChecksumReq req{sh.pool_id, sh.offset, sh.length};
+        auto resp = dc->call_raw(rpc::methods::DATA_CHECKSUM,
+                                 serde::to_bytes(req), 60000);
+        if (!resp.ok()) continue;  // transient worker trouble: retry next pass
+        ScrubU64Msg got;
+        if (!serde::from_bytes(resp.value(), got)) continue;
+        verified_any = true;
+        if (got.v != want) {
+          copy_bad = true;
+          bad_sh = &sh;
+          bad_want = want;
+          bad_got = got.v;
+          break;
+        }
+      }
+      if (!verified_any || !copy_bad) continue;
 
       // corrupt copy: quarantine it (re-validate placement under the lock)
       BB_LOG(ERROR) << "scrub: digest mismatch on " << cd.key << " @ "
-                    << sh.pool_id << "+" << sh.offset << " (want "
-                    << cd.checksum << " got " << got.v << ")";
+                    << bad_sh->pool_id << "+" << bad_sh->offset << " (want "
+                    << bad_want << " got " << bad_got << ")";
       bool dropped = false;
       bool object_gone = false;
       {

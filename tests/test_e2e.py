@@ -371,6 +371,43 @@ class TestScrub:
         finally:
             cl.stop()
 
+    def test_scrub_covers_striped_copies(self):
+        """Per-shard digests recorded at put time let the scrubber verify
+        STRIPED copies shard by shard (the round-1 scrubber silently skipped
+        them). Corrupting ONE shard of a 3-way striped object quarantines
+        the copy."""
+        cl = Cluster(n_workers=3, pool_bytes=32 * MB)
+        try:
+            c = cl.client()
+            cfg = bb.PlacementConfig()
+            cfg.max_workers_per_copy = 3
+            cfg.replication = 2
+            data = os.urandom(3 * MB)
+            c.put("striped-scrub", data, cfg)
+            ks = cl.keystone.service()
+            info = ks.get_workers("striped-scrub")
+            assert len(info.copies) == 2
+            assert len(info.copies[0].shards) == 3
+            # every shard carries its own standalone digest
+            for cp in info.copies:
+                for sh in cp.shards:
+                    assert sh.digest != 0
+            # clean pass first
+            assert ks.run_scrub_once() == 0
+            # corrupt the MIDDLE shard of copy 0 only
+            bad = info.copies[0].shards[1]
+            victim = next(w for w in cl.workers
+                          if any(p.pool_id == bad.pool_id
+                                 for p in w.pool_descriptors()))
+            victim.backend(bad.pool_id).write(bad.offset + 100, b"\xbe" * 64)
+            assert ks.run_scrub_once(64) == 1
+            info = ks.get_workers("striped-scrub")
+            assert len(info.copies) == 1  # corrupt striped copy quarantined
+            assert c.get("striped-scrub") == data  # survivor serves clean bytes
+            c.close()
+        finally:
+            cl.stop()
+
     def test_scrub_removes_fully_corrupt_object(self):
         cl = Cluster(n_workers=1, pool_bytes=16 * MB)
         try:
