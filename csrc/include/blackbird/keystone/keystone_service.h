@@ -198,9 +198,17 @@ class KeystoneService {
   std::map<ObjectKey, bool> dirty_;  // key → removed?
   std::condition_variable cv_;
   std::mutex cv_mu_;
+  // Small pool of connections per worker endpoint: the RPC server is
+  // thread-per-connection, so parallel migrations to one worker would
+  // serialize on a single shared connection's handler thread.
   rpc::RpcClient* data_client(const std::string& endpoint);
   std::mutex data_clients_mu_;
-  std::map<std::string, std::unique_ptr<rpc::RpcClient>> data_clients_;
+  static constexpr int kDataConns = 4;
+  struct DataConnPool {
+    std::vector<std::unique_ptr<rpc::RpcClient>> conns;
+    size_t cursor = 0;
+  };
+  std::map<std::string, DataConnPool> data_clients_;
   std::vector<uint64_t> watch_ids_;
   std::unique_ptr<coord::LeaderElector> elector_;
   std::string instance_id_;

@@ -755,13 +755,24 @@ void KeystoneService::keepalive_loop() {
 
 rpc::RpcClient* KeystoneService::data_client(const std::string& endpoint) {
   std::lock_guard<std::mutex> g(data_clients_mu_);
-  auto it = data_clients_.find(endpoint);
-  if (it != data_clients_.end() && it->second->connected())
-    return it->second.get();
-  auto c = std::make_unique<rpc::RpcClient>();
-  if (!c->connect(endpoint).ok()) return nullptr;
-  return data_clients_.insert_or_assign(endpoint, std::move(c))
-      .first->second.get();
+  auto& pool = data_clients_[endpoint];
+  // round-robin over up to kDataConns live connections (grown on demand);
+  // a dead connection is replaced in its slot
+  if (pool.conns.size() < kDataConns) {
+    auto c = std::make_unique<rpc::RpcClient>();
+    if (!c->connect(endpoint).ok())
+      return pool.conns.empty() ? nullptr : pool.conns[0].get();
+    pool.conns.push_back(std::move(c));
+    return pool.conns.back().get();
+  }
+  pool.cursor = (pool.cursor + 1) % pool.conns.size();
+  auto& slot = pool.conns[pool.cursor];
+  if (!slot->connected()) {
+    auto c = std::make_unique<rpc::RpcClient>();
+    if (!c->connect(endpoint).ok()) return nullptr;
+    slot = std::move(c);
+  }
+  return slot.get();
 }
 
 namespace {

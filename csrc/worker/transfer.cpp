@@ -1,5 +1,6 @@
 #include "blackbird/worker/transfer.h"
 
+#include <cstdlib>
 #include <cstring>
 
 #include <hip/hip_runtime_api.h>
@@ -35,8 +36,62 @@ struct ReadReq {
 TransferEngine::TransferEngine() : mapper_(std::make_shared<PoolMapper>()) {}
 
 TransferEngine::~TransferEngine() {
-  if (side_stream_) (void)hipStreamDestroy(static_cast<hipStream_t>(side_stream_));
-  if (staging_) (void)hipHostFree(staging_);
+  for (auto& l : lanes_) {
+    if (l.stream) (void)hipStreamDestroy(static_cast<hipStream_t>(l.stream));
+    if (l.pin) {
+      if (gpu::available()) (void)hipHostFree(l.pin);
+      else ::free(l.pin);
+    }
+  }
+}
+
+Result<TransferEngine::Lane*> TransferEngine::acquire_lane() {
+  std::unique_lock<std::mutex> lk(lanes_mu_);
+  Lane* l = nullptr;
+  lanes_cv_.wait(lk, [&] {
+    for (auto& cand : lanes_)
+      if (!cand.in_use) {
+        l = &cand;
+        return true;
+      }
+    return false;
+  });
+  l->in_use = true;
+  lk.unlock();
+  // lazy resource creation outside the pool lock (the lane is ours)
+  if (!l->pin) {
+    if (gpu::available()) {
+      void* p = nullptr;
+      if (hipHostMalloc(&p, 2 * kChunk, hipHostMallocDefault) != hipSuccess) {
+        release_lane(l);
+        return Error{ErrorCode::HIP_ERROR, "lane staging alloc"};
+      }
+      l->pin = p;
+    } else {
+      l->pin = ::malloc(2 * kChunk);
+      if (!l->pin) {
+        release_lane(l);
+        return Error{ErrorCode::INTERNAL_ERROR, "lane staging alloc"};
+      }
+    }
+  }
+  if (!l->stream && gpu::available()) {
+    hipStream_t s = nullptr;
+    if (hipStreamCreateWithFlags(&s, hipStreamNonBlocking) != hipSuccess) {
+      release_lane(l);
+      return Error{ErrorCode::HIP_ERROR, "lane stream create"};
+    }
+    l->stream = s;
+  }
+  return l;
+}
+
+void TransferEngine::release_lane(Lane* l) {
+  {
+    std::lock_guard<std::mutex> g(lanes_mu_);
+    l->in_use = false;
+  }
+  lanes_cv_.notify_one();
 }
 
 rpc::RpcClient* TransferEngine::data_client(const std::string& endpoint) {
@@ -51,16 +106,21 @@ rpc::RpcClient* TransferEngine::data_client(const std::string& endpoint) {
 Result<void> TransferEngine::pull(StorageBackend& dst, uint64_t dst_offset,
                                   const std::vector<ShardPlacement>& srcs) {
   BB_TRACE_SCOPE("bb::tier_pull");
+  auto lane = acquire_lane();
+  if (!lane.ok()) return lane.error();
   uint64_t off = dst_offset;
+  Result<void> rc{};
   for (const auto& s : srcs) {
-    BB_RETURN_IF_ERROR(pull_one(dst, off, s));
+    rc = pull_one(dst, off, s, *lane.value());
+    if (!rc.ok()) break;
     off += s.length;
   }
-  return {};
+  release_lane(lane.value());
+  return rc;
 }
 
 Result<void> TransferEngine::pull_one(StorageBackend& dst, uint64_t dst_offset,
-                                      const ShardPlacement& src) {
+                                      const ShardPlacement& src, Lane& lane) {
   const bool dst_is_gpu = dst.storage_class() == StorageClass::RAM_GPU;
   uint8_t* dst_ptr = static_cast<uint8_t*>(dst.base_ptr());
 
@@ -101,17 +161,32 @@ Result<void> TransferEngine::pull_one(StorageBackend& dst, uint64_t dst_offset,
 
   if (src_ptr && !dst_ptr) {
     // destination has no memory mapping (direct-IO file tier): move through
-    // backend write(); GPU sources stage via a pinned bounce first
+    // backend write(). Host sources write straight through; GPU sources
+    // ping-pong the lane's two pinned halves so the D2H DMA of chunk i+1
+    // overlaps the (CPU/NVMe) backend write of chunk i.
     if (!src_is_gpu) return dst.write(dst_offset, src_ptr, src.length);
-    if (!staging_) {
-      BB_HIP(hipHostMalloc(&staging_, staging_size_, hipHostMallocDefault));
-    }
-    uint64_t done = 0;
-    while (done < src.length) {
-      uint64_t chunk = std::min(src.length - done, staging_size_);
-      BB_HIP(hipMemcpy(staging_, src_ptr + done, chunk, hipMemcpyDeviceToHost));
-      BB_RETURN_IF_ERROR(dst.write(dst_offset + done, staging_, chunk));
-      done += chunk;
+    auto stream = static_cast<hipStream_t>(lane.stream);
+    uint8_t* half[2] = {static_cast<uint8_t*>(lane.pin),
+                        static_cast<uint8_t*>(lane.pin) + kChunk};
+    const uint64_t n_chunks = (src.length + kChunk - 1) / kChunk;
+    uint64_t cur_len = std::min(src.length, kChunk);
+    BB_HIP(hipMemcpyAsync(half[0], src_ptr, cur_len, hipMemcpyDeviceToHost,
+                          stream));
+    BB_HIP(hipStreamSynchronize(stream));
+    for (uint64_t c = 0; c < n_chunks; ++c) {
+      const int cur = static_cast<int>(c & 1);
+      uint64_t next_off = (c + 1) * kChunk;
+      uint64_t next_len = 0;
+      if (c + 1 < n_chunks) {
+        next_len = std::min(src.length - next_off, kChunk);
+        BB_HIP(hipMemcpyAsync(half[1 - cur], src_ptr + next_off, next_len,
+                              hipMemcpyDeviceToHost, stream));
+      }
+      BB_RETURN_IF_ERROR(dst.write(dst_offset + c * kChunk, half[cur], cur_len));
+      if (c + 1 < n_chunks) {
+        BB_HIP(hipStreamSynchronize(stream));
+        cur_len = next_len;
+      }
     }
     return {};
   }
@@ -120,14 +195,10 @@ Result<void> TransferEngine::pull_one(StorageBackend& dst, uint64_t dst_offset,
       std::memcpy(dst_ptr + dst_offset, src_ptr, src.length);
       return {};
     }
-    // at least one GPU endpoint: hipMemcpyAsync on the side stream, gated by
-    // stream sync (PINNED_CPU destinations take the DMA fast path).
-    if (!side_stream_) {
-      hipStream_t s = nullptr;
-      BB_HIP(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
-      side_stream_ = s;
-    }
-    auto stream = static_cast<hipStream_t>(side_stream_);
+    // at least one GPU endpoint: hipMemcpyAsync on the LANE's stream
+    // (PINNED_CPU destinations take the DMA fast path); concurrent pulls
+    // ride different lanes/streams, so the engines overlap
+    auto stream = static_cast<hipStream_t>(lane.stream);
     hipMemcpyKind kind = src_is_gpu
                              ? (dst_is_gpu ? hipMemcpyDeviceToDevice
                                            : hipMemcpyDeviceToHost)
@@ -137,23 +208,15 @@ Result<void> TransferEngine::pull_one(StorageBackend& dst, uint64_t dst_offset,
     return {};
   }
 
-  // TCP fallback: read from the source worker's data plane into staging,
-  // then backend write (handles GPU destinations internally).
+  // TCP fallback: read from the source worker's data plane into the lane's
+  // staging halves, then backend write (handles GPU destinations internally).
   auto* dc = data_client(src.access.endpoint);
   if (!dc)
     return Error{ErrorCode::CONNECT_FAILED,
                  "pull: data plane " + src.access.endpoint};
-  if (!staging_) {
-    if (gpu::available()) {
-      BB_HIP(hipHostMalloc(&staging_, staging_size_, hipHostMallocDefault));
-    } else {
-      staging_ = malloc(staging_size_);
-      if (!staging_) return Error{ErrorCode::INTERNAL_ERROR, "staging alloc"};
-    }
-  }
   uint64_t done = 0;
   while (done < src.length) {
-    uint64_t chunk = std::min(src.length - done, staging_size_);
+    uint64_t chunk = std::min(src.length - done, 2 * kChunk);
     ReadReq req{src.pool_id, src.offset + done, chunk};
     auto r = dc->call_raw(rpc::methods::DATA_READ, serde::to_bytes(req));
     if (!r.ok()) return r.error();
