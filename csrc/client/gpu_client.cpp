@@ -79,10 +79,16 @@ Result<void> GpuClient::init() {
 GpuClient::Resolved GpuClient::resolve_device_ptr(const ShardPlacement& s) {
   bool is_dev = false;
   int dev = -1;
-  if (void* base = LocalPools::inst().lookup(s.pool_id, &is_dev, &dev)) {
+  uint64_t pool_size = 0;
+  if (void* base = LocalPools::inst().lookup(s.pool_id, &is_dev, &dev,
+                                             &pool_size)) {
     if (is_dev)
       return {static_cast<uint8_t*>(base) + s.offset, dev == device_};
-    return {};  // host pool: use staged path
+    // host pool in this process (pinned/shm tier): pin+map once so the GPU
+    // addresses it directly (PCIe DMA) instead of staging through a bounce
+    if (void* dm = c_.mapper_->host_dev_map(s.pool_id, base, pool_size))
+      return {static_cast<uint8_t*>(dm) + s.offset, false};
+    return {};  // unmappable host pool: staged path
   }
   AccessInfo a = s.access;
   if (a.endpoint.empty()) {
@@ -90,11 +96,50 @@ GpuClient::Resolved GpuClient::resolve_device_ptr(const ShardPlacement& s) {
     if (!r.ok()) return {};
     a = std::move(r.value());
   }
+  if (a.kind == AccessKind::SHM && !a.shm_name.empty()) {
+    // host pool of ANOTHER process: shm-map it, then GPU-map the mapping
+    uint64_t sz = 0;
+    if (void* base = c_.mapper_->map_shm(a.shm_name, 0, &sz))
+      if (void* dm = c_.mapper_->host_dev_map(s.pool_id, base, sz))
+        return {static_cast<uint8_t*>(dm) + s.offset, false};
+    return {};
+  }
   if (a.kind != AccessKind::HIP_IPC || a.ipc_handle_hex.empty()) return {};
   void* base = c_.mapper_->open_ipc(a.ipc_handle_hex, a.device_id);
   if (!base) return {};
   // IPC pools come from other processes (one rank per GPU) ⇒ cross-device
   return {static_cast<uint8_t*>(base) + s.offset, false};
+}
+
+// Shared pool-table resolution for the v2 batch protocols: device-visible
+// base (HBM local, IPC peer, or GPU-mapped host tier) or nullptr (staged).
+uint8_t* GpuClient::resolve_pool_base(const PoolId& pool_id, AccessInfo* access,
+                                      bool* same_device) {
+  bool is_dev = false;
+  int dev = -1;
+  uint64_t pool_size = 0;
+  if (void* base = LocalPools::inst().lookup(pool_id, &is_dev, &dev,
+                                             &pool_size)) {
+    if (is_dev) {
+      if (same_device) *same_device = dev == device_;
+      return static_cast<uint8_t*>(base);
+    }
+    return static_cast<uint8_t*>(
+        c_.mapper_->host_dev_map(pool_id, base, pool_size));
+  }
+  auto a = c_.pool_access(pool_id);
+  if (!a.ok()) return nullptr;
+  uint8_t* out = nullptr;
+  if (a->kind == AccessKind::HIP_IPC && !a->ipc_handle_hex.empty()) {
+    out = static_cast<uint8_t*>(
+        c_.mapper_->open_ipc(a->ipc_handle_hex, a->device_id));
+  } else if (a->kind == AccessKind::SHM && !a->shm_name.empty()) {
+    uint64_t sz = 0;
+    if (void* base = c_.mapper_->map_shm(a->shm_name, 0, &sz))
+      out = static_cast<uint8_t*>(c_.mapper_->host_dev_map(pool_id, base, sz));
+  }
+  if (access) *access = std::move(a.value());
+  return out;
 }
 
 Result<void> GpuClient::staged_write(const ShardPlacement& s, const void* dev_src) {
@@ -335,24 +380,8 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
   std::vector<PoolRef> pools(npools);
   for (uint16_t i = 0; i < npools; ++i) {
     pools[i].pool_id = d.str();
-    bool is_dev = false;
-    int dev = -1;
-    if (void* base = LocalPools::inst().lookup(pools[i].pool_id, &is_dev, &dev)) {
-      if (is_dev) {
-        pools[i].base = static_cast<uint8_t*>(base);
-        pools[i].same_device = dev == device_;
-      }
-      continue;
-    }
-    auto a = c_.pool_access(pools[i].pool_id);
-    if (!a.ok()) continue;
-    pools[i].access = std::move(a.value());
-    if (pools[i].access.kind == AccessKind::HIP_IPC &&
-        !pools[i].access.ipc_handle_hex.empty()) {
-      if (void* base = c_.mapper_->open_ipc(pools[i].access.ipc_handle_hex,
-                                            pools[i].access.device_id))
-        pools[i].base = static_cast<uint8_t*>(base);
-    }
+    pools[i].base = resolve_pool_base(pools[i].pool_id, &pools[i].access,
+                                      &pools[i].same_device);
   }
 
   std::vector<int32_t> statuses(items.size(), 0);
@@ -526,16 +555,7 @@ void GpuClient::clear_placement_cache() {
 }
 
 uint8_t* GpuClient::device_pool_base(const PoolId& id) {
-  bool is_dev = false;
-  int dev = -1;
-  if (void* b = LocalPools::inst().lookup(id, &is_dev, &dev))
-    return is_dev ? static_cast<uint8_t*>(b) : nullptr;
-  auto a = c_.pool_access(id);
-  if (!a.ok()) return nullptr;
-  if (a->kind == AccessKind::HIP_IPC && !a->ipc_handle_hex.empty())
-    return static_cast<uint8_t*>(
-        c_.mapper_->open_ipc(a->ipc_handle_hex, a->device_id));
-  return nullptr;
+  return resolve_pool_base(id, nullptr, nullptr);
 }
 
 // Session fast path for gets: descs and want-digest slots were resolved on a
@@ -698,24 +718,8 @@ Result<std::vector<int32_t>> GpuClient::batch_get_device_rpc(
   std::vector<PoolRef> pools(npools);
   for (uint16_t i = 0; i < npools; ++i) {
     pools[i].pool_id = d.str();
-    bool is_dev = false;
-    int dev = -1;
-    if (void* base = LocalPools::inst().lookup(pools[i].pool_id, &is_dev, &dev)) {
-      if (is_dev) {
-        pools[i].base = static_cast<uint8_t*>(base);
-        pools[i].same_device = dev == device_;
-      }
-      continue;
-    }
-    auto a = c_.pool_access(pools[i].pool_id);
-    if (!a.ok()) continue;
-    pools[i].access = std::move(a.value());
-    if (pools[i].access.kind == AccessKind::HIP_IPC &&
-        !pools[i].access.ipc_handle_hex.empty()) {
-      if (void* base = c_.mapper_->open_ipc(pools[i].access.ipc_handle_hex,
-                                            pools[i].access.device_id))
-        pools[i].base = static_cast<uint8_t*>(base);
-    }
+    pools[i].base = resolve_pool_base(pools[i].pool_id, &pools[i].access,
+                                      &pools[i].same_device);
   }
 
   std::vector<int32_t> statuses(items.size(), 0);

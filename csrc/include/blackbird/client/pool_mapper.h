@@ -11,6 +11,7 @@
 #include <mutex>
 #include <set>
 #include <string>
+#include <vector>
 
 #include <hip/hip_runtime_api.h>
 
@@ -21,6 +22,7 @@ namespace blackbird {
 class PoolMapper {
  public:
   ~PoolMapper() {
+    for (void* p : registered_) (void)hipHostUnregister(p);
     for (auto& [k, m] : shm_) munmap(m.ptr, m.size);
     for (auto& [k, p] : ipc_) (void)hipIpcCloseMemHandle(p);
   }
@@ -53,6 +55,33 @@ class PoolMapper {
     return p;
   }
 
+  // Device-visible mapping of a HOST pool: pins (hipHostRegister, unless the
+  // memory is already page-locked) and maps it into the GPU address space
+  // once, cached per pool. Lets the fused copy/digest kernels read and write
+  // host tiers (PINNED_CPU, local RAM_CPU/shm) directly at PCIe DMA speed
+  // instead of double-copying through a staging bounce. nullptr when the GPU
+  // cannot address it (registration failed / no GPU).
+  void* host_dev_map(const std::string& pool_key, void* base, uint64_t size) {
+    std::lock_guard<std::mutex> g(mu_);
+    auto it = dev_maps_.find(pool_key);
+    if (it != dev_maps_.end()) return it->second;
+    if (dev_map_failed_.count(pool_key)) return nullptr;
+    void* dp = nullptr;
+    if (hipHostGetDevicePointer(&dp, base, 0) != hipSuccess) {
+      if (hipHostRegister(base, size, hipHostRegisterDefault) != hipSuccess) {
+        dev_map_failed_.insert(pool_key);
+        return nullptr;
+      }
+      registered_.push_back(base);
+      if (hipHostGetDevicePointer(&dp, base, 0) != hipSuccess) {
+        dev_map_failed_.insert(pool_key);
+        return nullptr;
+      }
+    }
+    dev_maps_[pool_key] = dp;
+    return dp;
+  }
+
   // HIP IPC: returns device pointer valid in this process, or nullptr.
   // Negative cache: a handle that failed to open is not retried (handles are
   // immutable per allocation).
@@ -81,6 +110,9 @@ class PoolMapper {
   std::map<std::string, Shm> shm_;
   std::map<std::string, void*> ipc_;
   std::set<std::string> ipc_failed_;
+  std::map<std::string, void*> dev_maps_;  // pool key → device-visible ptr
+  std::set<std::string> dev_map_failed_;
+  std::vector<void*> registered_;  // bases WE pinned (unregister on teardown)
 };
 
 }  // namespace blackbird
