@@ -175,6 +175,52 @@ Result<void> GpuClient::staged_read_buf(const ShardPlacement& s, void* dev_dst,
   return {};
 }
 
+Result<void> GpuClient::staged_write_buf(const ShardPlacement& s,
+                                         const void* dev_src, void* staging,
+                                         uint64_t staging_size) {
+  uint64_t done = 0;
+  while (done < s.length) {
+    uint64_t chunk = std::min(s.length - done, staging_size);
+    BB_HIP(hipMemcpy(staging, static_cast<const uint8_t*>(dev_src) + done,
+                     chunk, hipMemcpyDeviceToHost));
+    ShardPlacement part = s;
+    part.offset = s.offset + done;
+    part.length = chunk;
+    BB_RETURN_IF_ERROR(c_.write_shard(part, staging));
+    done += chunk;
+  }
+  return {};
+}
+
+Result<void> GpuClient::staged_write_many(
+    const std::vector<std::pair<ShardPlacement, const void*>>& work) {
+  if (work.empty()) return {};
+  if (work.size() == 1) return staged_write(work[0].first, work[0].second);
+  const int nthreads = std::min<int>(4, static_cast<int>(work.size()));
+  std::atomic<size_t> next{0};
+  std::vector<std::future<Result<void>>> futs;
+  const uint64_t per_buf = 16ull << 20;
+  for (int t = 0; t < nthreads; ++t)
+    futs.push_back(std::async(std::launch::async, [&]() -> Result<void> {
+      void* buf = nullptr;
+      if (hipHostMalloc(&buf, per_buf, hipHostMallocDefault) != hipSuccess)
+        return Error{ErrorCode::HIP_ERROR, "staging alloc"};
+      Result<void> rc{};
+      for (size_t i = next.fetch_add(1); i < work.size();
+           i = next.fetch_add(1)) {
+        auto r = staged_write_buf(work[i].first, work[i].second, buf, per_buf);
+        if (!r.ok()) {
+          rc = r;
+          break;
+        }
+      }
+      (void)hipHostFree(buf);
+      return rc;
+    }));
+  for (auto& f : futs) BB_RETURN_IF_ERROR(f.get());
+  return {};
+}
+
 Result<void> GpuClient::staged_read(const ShardPlacement& s, void* dev_dst) {
   std::lock_guard<std::mutex> g(staging_mu_);  // async batches share staging_
   return staged_read_buf(s, dev_dst, staging_, staging_size_);
@@ -390,6 +436,8 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
   std::vector<uint32_t> fused_hash_idx;
   std::vector<std::pair<PoolId, uint64_t>> fused_hash_loc;  // placement cache
   std::vector<uint32_t> committed_idx;
+  std::vector<std::pair<ShardPlacement, const void*>> staged_put_work;
+  std::vector<uint32_t> staged_put_idx;
   int si = 0;
 
   for (size_t i = 0; i < items.size() && d.ok(); ++i) {
@@ -429,8 +477,8 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
         sp.offset = off;
         sp.length = items[i].size;
         sp.access = pr.access;
-        auto r = staged_write(sp, src);
-        if (!r.ok()) { ok = false; break; }
+        staged_put_work.emplace_back(std::move(sp), src);
+        staged_put_idx.push_back(static_cast<uint32_t>(i));
       }
     }
     if (!ok) {
@@ -441,6 +489,22 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
     else committed_idx.push_back(static_cast<uint32_t>(i));
   }
   if (!d.ok()) return Error{ErrorCode::PROTOCOL_ERROR, "bad v2 response"};
+  if (!staged_put_work.empty()) {
+    // unmappable pools (direct-IO NVMe tier, TCP-only remotes): fan the
+    // staged writes out instead of one sync'd bounce per object
+    auto r = staged_write_many(staged_put_work);
+    if (!r.ok()) {
+      for (auto i : staged_put_idx)
+        statuses[i] = static_cast<int32_t>(r.code());
+      // drop failed items from the commit set (staged work only ever comes
+      // from non-fused items, so fused_hash_idx is unaffected; this happens
+      // BEFORE digests are computed, keeping digests[] aligned)
+      std::vector<uint32_t> keep;
+      for (auto i : committed_idx)
+        if (statuses[i] == 0) keep.push_back(i);
+      committed_idx.swap(keep);
+    }
+  }
 
   if (!fused.empty()) {
     auto r = gpu::batched_copy(fused.data(), static_cast<uint32_t>(fused.size()),

@@ -157,6 +157,11 @@ class GpuClient {
   // device-visible base of a pool (local or IPC-mapped), or nullptr
   uint8_t* device_pool_base(const PoolId& id);
   Result<void> staged_write(const ShardPlacement& s, const void* dev_src);
+  Result<void> staged_write_buf(const ShardPlacement& s, const void* dev_src,
+                                void* staging, uint64_t staging_size);
+  // fan staged writes (direct-IO/TCP pools) out over pinned buffers
+  Result<void> staged_write_many(
+      const std::vector<std::pair<ShardPlacement, const void*>>& work);
   Result<void> staged_read(const ShardPlacement& s, void* dev_dst);
   Result<void> staged_read_buf(const ShardPlacement& s, void* dev_dst,
                                void* staging, uint64_t staging_size);
