@@ -234,6 +234,10 @@ Result<void> Client::write_shard(const ShardPlacement& s, const void* src) {
       }
       if (hipMemcpy(dst, src, s.length, hipMemcpyHostToDevice) == hipSuccess)
         return {};
+    } else if (auto* be = LocalPools::inst().backend(s.pool_id)) {
+      // unmapped same-process pool (direct-IO NVMe tier): call the backend
+      // directly instead of framing the bytes through TCP loopback
+      return be->write(s.offset, src, s.length);
     }
   }
   if (a.kind == AccessKind::SHM && !a.shm_name.empty()) {
@@ -304,6 +308,10 @@ Result<void> Client::read_shard(const ShardPlacement& s, void* dst) {
       }
       if (hipMemcpy(dst, src2, s.length, hipMemcpyDeviceToHost) == hipSuccess)
         return {};
+    } else if (auto* be = LocalPools::inst().backend(s.pool_id)) {
+      // unmapped same-process pool (direct-IO NVMe tier): read through the
+      // backend instead of TCP loopback
+      return be->read(s.offset, dst, s.length);
     }
   }
   if (a.kind == AccessKind::SHM && !a.shm_name.empty()) {

@@ -103,11 +103,15 @@ class LocalPools {
  public:
   static LocalPools& inst();
   void add(const PoolId& id, void* base, uint64_t size, bool is_device,
-           int device);
+           int device, StorageBackend* backend = nullptr);
   void remove(const PoolId& id);
   // returns base or nullptr; *is_device/*device/*size set when found
   void* lookup(const PoolId& id, bool* is_device = nullptr,
                int* device = nullptr, uint64_t* size = nullptr);
+  // same-process backend handle (valid while the pool is registered): lets
+  // clients embedded in the worker process reach UNMAPPED tiers (direct-IO
+  // NVMe) through the backend's read/write instead of TCP loopback
+  StorageBackend* backend(const PoolId& id);
 
  private:
   struct Entry {
@@ -115,6 +119,7 @@ class LocalPools {
     uint64_t size;
     bool is_device;
     int device;
+    StorageBackend* backend;
   };
   std::mutex mu_;
   std::map<PoolId, Entry> pools_;

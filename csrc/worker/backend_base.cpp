@@ -114,9 +114,15 @@ LocalPools& LocalPools::inst() {
 }
 
 void LocalPools::add(const PoolId& id, void* base, uint64_t size,
-                     bool is_device, int device) {
+                     bool is_device, int device, StorageBackend* backend) {
   std::lock_guard<std::mutex> g(mu_);
-  pools_[id] = Entry{base, size, is_device, device};
+  pools_[id] = Entry{base, size, is_device, device, backend};
+}
+
+StorageBackend* LocalPools::backend(const PoolId& id) {
+  std::lock_guard<std::mutex> g(mu_);
+  auto it = pools_.find(id);
+  return it == pools_.end() ? nullptr : it->second.backend;
 }
 
 void LocalPools::remove(const PoolId& id) {

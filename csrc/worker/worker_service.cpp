@@ -90,7 +90,7 @@ Result<void> WorkerService::initialize() {
     LocalPools::inst().add(pc.pool_id, b.value()->base_ptr(),
                            b.value()->capacity(),
                            pc.storage_class == StorageClass::RAM_GPU,
-                           pc.gpu_device_id);
+                           pc.gpu_device_id, b.value().get());
     backends_[pc.pool_id] = std::move(b.value());
   }
   auto hp = net::split_endpoint(config_.data_listen_address);
