@@ -489,22 +489,15 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
     else committed_idx.push_back(static_cast<uint32_t>(i));
   }
   if (!d.ok()) return Error{ErrorCode::PROTOCOL_ERROR, "bad v2 response"};
-  if (!staged_put_work.empty()) {
-    // unmappable pools (direct-IO NVMe tier, TCP-only remotes): fan the
-    // staged writes out instead of one sync'd bounce per object
-    auto r = staged_write_many(staged_put_work);
-    if (!r.ok()) {
-      for (auto i : staged_put_idx)
-        statuses[i] = static_cast<int32_t>(r.code());
-      // drop failed items from the commit set (staged work only ever comes
-      // from non-fused items, so fused_hash_idx is unaffected; this happens
-      // BEFORE digests are computed, keeping digests[] aligned)
-      std::vector<uint32_t> keep;
-      for (auto i : committed_idx)
-        if (statuses[i] == 0) keep.push_back(i);
-      committed_idx.swap(keep);
-    }
-  }
+  // unmappable pools (direct-IO NVMe tier, TCP-only remotes): fan the staged
+  // writes out on a side thread so the disk overlaps the kernel launches and
+  // digest passes below
+  std::future<Result<void>> staged_fut;
+  if (!staged_put_work.empty())
+    staged_fut = std::async(std::launch::async, [&] {
+      (void)hipSetDevice(device_);
+      return staged_write_many(staged_put_work);
+    });
 
   if (!fused.empty()) {
     auto r = gpu::batched_copy(fused.data(), static_cast<uint32_t>(fused.size()),
@@ -537,11 +530,19 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
   if (!fused.empty()) BB_HIP(hipStreamSynchronize(streams_[0]));
   for (int j = 0; j < std::min(si, kStreams); ++j)
     BB_HIP(hipStreamSynchronize(streams_[j]));
+  if (staged_fut.valid()) {
+    auto r = staged_fut.get();
+    if (!r.ok())  // failed staged items must not commit (cancelled below)
+      for (auto i : staged_put_idx)
+        statuses[i] = static_cast<int32_t>(ErrorCode::TRANSFER_FAILED);
+  }
 
   PutCompleteListMsg completes;
-  for (size_t j = 0; j < committed_idx.size(); ++j)
+  for (size_t j = 0; j < committed_idx.size(); ++j) {
+    if (statuses[committed_idx[j]] != 0) continue;  // staged write failed
     completes.reqs.push_back(
         PutCompleteRequest{items[committed_idx[j]].key, digests[j]});
+  }
   for (size_t j = 0; j < fused_hash_idx.size(); ++j)
     completes.reqs.push_back(
         PutCompleteRequest{items[fused_hash_idx[j]].key, fused_digests[j]});
