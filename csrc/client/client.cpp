@@ -134,11 +134,14 @@ Result<std::string> Client::meta_call_raw(uint16_t m, const std::string& body,
                                           int timeout_ms) {
   if (timeout_ms <= 0) timeout_ms = opts_.rpc_timeout_ms;
   auto r = meta_.call_raw(m, body, timeout_ms);
-  for (int attempt = 0; attempt < 8 && !r.ok(); ++attempt) {
+  // ~6 s total budget: leader election + registry TTL convergence after an
+  // unclean leader death take a few seconds
+  for (int attempt = 0; attempt < 20 && !r.ok(); ++attempt) {
     const bool standby = r.code() == ErrorCode::NOT_LEADER;
     switch (r.code()) {
       case ErrorCode::NOT_LEADER:
       case ErrorCode::NOT_CONNECTED:
+      case ErrorCode::CONNECT_FAILED:
       case ErrorCode::CONNECTION_CLOSED:
       case ErrorCode::SEND_FAILED:
       case ErrorCode::RECV_FAILED:
@@ -148,15 +151,21 @@ Result<std::string> Client::meta_call_raw(uint16_t m, const std::string& body,
     }
     // a standby answer with no discovery path configured cannot improve
     if (standby && opts_.coord_endpoint.empty()) return r;
-    std::this_thread::sleep_for(
-        std::chrono::milliseconds(standby ? 300 : 50));
+    std::this_thread::sleep_for(std::chrono::milliseconds(
+        std::min(50 * (attempt + 1), standby ? 500 : 400)));
     {
       std::lock_guard<std::mutex> g(reconnect_mu_);
       if (standby || !meta_.connected()) {
         if (!opts_.coord_endpoint.empty())
           opts_.keystone_endpoint.clear();  // force re-discovery
         auto rc = connect();
-        if (!rc.ok()) continue;
+        if (!rc.ok()) {
+          // a dead (or not-yet-elected) leader: keep re-discovering — the
+          // registry TTL drops the dead instance and the standby registers
+          if (!opts_.coord_endpoint.empty()) opts_.keystone_endpoint.clear();
+          continue;
+        }
+        reconnect_gen_.fetch_add(1);
       }
     }
     r = meta_.call_raw(m, body, timeout_ms);
@@ -431,6 +440,18 @@ std::vector<std::vector<uint64_t>> host_shard_digests(
 
 Result<void> Client::put(const ObjectKey& key, const void* data, uint64_t size,
                          const PlacementConfig& cfg) {
+  const uint64_t gen = reconnect_gen_.load();
+  auto r = put_once(key, data, size, cfg);
+  if (r.ok() || reconnect_gen_.load() == gen ||
+      !failover_retriable(static_cast<int32_t>(r.code())))
+    return r;
+  // leader change mid-put: the new leader never saw the PENDING object —
+  // redo once from put_start
+  return put_once(key, data, size, cfg);
+}
+
+Result<void> Client::put_once(const ObjectKey& key, const void* data,
+                              uint64_t size, const PlacementConfig& cfg) {
   PutStartRequest req{key, size, cfg};
   auto start = meta_call<PutStartRequest, PutStartResponse>(M::PUT_START, req,
                                                              opts_.rpc_timeout_ms);
@@ -511,7 +532,46 @@ Result<std::vector<int32_t>> Client::batch_remove(
 
 
 
+bool Client::failover_retriable(int32_t st) {
+  switch (static_cast<ErrorCode>(st)) {
+    case ErrorCode::OBJECT_NOT_FOUND:   // new leader never saw the PENDING put
+    case ErrorCode::NOT_LEADER:
+    case ErrorCode::NOT_CONNECTED:
+    case ErrorCode::CONNECT_FAILED:
+    case ErrorCode::CONNECTION_CLOSED:
+    case ErrorCode::SEND_FAILED:
+    case ErrorCode::RECV_FAILED:
+    case ErrorCode::RPC_FAILED:
+      return true;
+    default:
+      return false;
+  }
+}
+
 Result<std::vector<int32_t>> Client::batch_put(const std::vector<PutItem>& items,
+                                               const PlacementConfig& cfg) {
+  const uint64_t gen = reconnect_gen_.load();
+  auto st = batch_put_once(items, cfg);
+  if (!st.ok()) return st;
+  if (reconnect_gen_.load() == gen) return st;  // no failover: statuses final
+  // a leader change happened during the batch: redo the items whose failure
+  // is failover-shaped (the new leader lost their PENDING state) exactly once
+  std::vector<PutItem> redo;
+  std::vector<size_t> redo_idx;
+  for (size_t i = 0; i < items.size(); ++i)
+    if (failover_retriable(st.value()[i])) {
+      redo.push_back(items[i]);
+      redo_idx.push_back(i);
+    }
+  if (redo.empty()) return st;
+  auto st2 = batch_put_once(redo, cfg);
+  if (!st2.ok()) return st;  // keep the first answer
+  for (size_t j = 0; j < redo_idx.size(); ++j)
+    st.value()[redo_idx[j]] = st2.value()[j];
+  return st;
+}
+
+Result<std::vector<int32_t>> Client::batch_put_once(const std::vector<PutItem>& items,
                                                const PlacementConfig& cfg) {
   BatchPutStartRequest breq;
   breq.requests.reserve(items.size());
@@ -569,6 +629,7 @@ Result<std::vector<int32_t>> Client::batch_put(const std::vector<PutItem>& items
             digests[i] = gpu::checksum_cpu(items[i].data, items[i].size);
       PutCompleteListMsg completes2;
       std::vector<std::string> cancels2;
+      std::vector<size_t> complete_idx2;
       for (size_t i = 0; i < items.size(); ++i) {
         if (start->items[i].status != 0) continue;
         if (statuses[i] != 0) {
@@ -580,11 +641,17 @@ Result<std::vector<int32_t>> Client::batch_put(const std::vector<PutItem>& items
           pc.shard_digests =
               host_shard_digests(start->items[i].copies, items[i].data);
         completes2.reqs.push_back(std::move(pc));
+        complete_idx2.push_back(i);
       }
       if (!completes2.reqs.empty()) {
         auto r = meta_call<PutCompleteListMsg, StatusListMsg>(
             M::BATCH_PUT_COMPLETE, completes2, opts_.rpc_timeout_ms);
         if (!r.ok()) return r.error();
+        // per-item commit failures (e.g. a failover lost the PENDING state)
+        // must reach the caller, not vanish
+        for (size_t j = 0; j < complete_idx2.size() &&
+                           j < r->statuses.size(); ++j)
+          if (r->statuses[j] != 0) statuses[complete_idx2[j]] = r->statuses[j];
       }
       if (!cancels2.empty())
         meta_call_raw(M::BATCH_PUT_CANCEL, serde::to_bytes(KeysMsg{cancels2}),
@@ -621,6 +688,7 @@ Result<std::vector<int32_t>> Client::batch_put(const std::vector<PutItem>& items
   }
   PutCompleteListMsg completes;
   std::vector<std::string> cancels;
+  std::vector<size_t> complete_idx;
   for (size_t i = 0; i < items.size(); ++i) {
     if (start->items[i].status != 0) continue;  // placement failed
     if (statuses[i] != 0) {
@@ -632,11 +700,14 @@ Result<std::vector<int32_t>> Client::batch_put(const std::vector<PutItem>& items
       pc.shard_digests =
           host_shard_digests(start->items[i].copies, items[i].data);
     completes.reqs.push_back(std::move(pc));
+    complete_idx.push_back(i);
   }
   if (!completes.reqs.empty()) {
     auto r = meta_call<PutCompleteListMsg, StatusListMsg>(
         M::BATCH_PUT_COMPLETE, completes, opts_.rpc_timeout_ms);
     if (!r.ok()) return r.error();
+    for (size_t j = 0; j < complete_idx.size() && j < r->statuses.size(); ++j)
+      if (r->statuses[j] != 0) statuses[complete_idx[j]] = r->statuses[j];
   }
   if (!cancels.empty())
     meta_call_raw(M::BATCH_PUT_CANCEL, serde::to_bytes(KeysMsg{cancels}),
@@ -645,6 +716,28 @@ Result<std::vector<int32_t>> Client::batch_put(const std::vector<PutItem>& items
 }
 
 Result<std::vector<std::pair<int32_t, std::string>>> Client::batch_get(
+    const std::vector<ObjectKey>& keys) {
+  const uint64_t gen = reconnect_gen_.load();
+  auto out = batch_get_once(keys);
+  if (!out.ok() || reconnect_gen_.load() == gen) return out;
+  // failover mid-batch: the new leader may briefly lag the persistence
+  // stream — redo failover-shaped per-item failures once
+  std::vector<ObjectKey> redo;
+  std::vector<size_t> redo_idx;
+  for (size_t i = 0; i < keys.size(); ++i)
+    if (failover_retriable(out.value()[i].first)) {
+      redo.push_back(keys[i]);
+      redo_idx.push_back(i);
+    }
+  if (redo.empty()) return out;
+  auto out2 = batch_get_once(redo);
+  if (!out2.ok()) return out;
+  for (size_t j = 0; j < redo_idx.size(); ++j)
+    out.value()[redo_idx[j]] = std::move(out2.value()[j]);
+  return out;
+}
+
+Result<std::vector<std::pair<int32_t, std::string>>> Client::batch_get_once(
     const std::vector<ObjectKey>& keys) {
   auto meta = meta_call<KeysMsg, BatchGetWorkersResponse>(
       M::BATCH_GET_WORKERS, KeysMsg{keys}, opts_.rpc_timeout_ms);

@@ -914,6 +914,29 @@ Result<std::vector<int32_t>> GpuClient::batch_get_device_rpc(
 Result<std::vector<int32_t>> GpuClient::batch_put_device(
     const std::vector<DevPutItem>& items, const PlacementConfig& cfg,
     BatchPutSession* sess) {
+  const uint64_t gen = c_.reconnect_generation();
+  auto st = batch_put_device_once(items, cfg, sess);
+  if (!st.ok() || c_.reconnect_generation() == gen) return st;
+  // keystone failover mid-batch: the new leader never saw this batch's
+  // PENDING state — redo the failover-shaped failures once
+  std::vector<DevPutItem> redo;
+  std::vector<size_t> redo_idx;
+  for (size_t i = 0; i < items.size(); ++i)
+    if (Client::failover_retriable(st.value()[i])) {
+      redo.push_back(items[i]);
+      redo_idx.push_back(i);
+    }
+  if (redo.empty()) return st;
+  auto st2 = batch_put_device_once(redo, cfg, nullptr);
+  if (!st2.ok()) return st;
+  for (size_t j = 0; j < redo_idx.size(); ++j)
+    st.value()[redo_idx[j]] = st2.value()[j];
+  return st;
+}
+
+Result<std::vector<int32_t>> GpuClient::batch_put_device_once(
+    const std::vector<DevPutItem>& items, const PlacementConfig& cfg,
+    BatchPutSession* sess) {
   BB_RETURN_IF_ERROR(init());
   BB_HIP(hipSetDevice(device_));
   if (cfg.max_workers_per_copy <= 1)
@@ -1059,19 +1082,25 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device(
   }
 
   PutCompleteListMsg completes;
+  std::vector<uint32_t> complete_order;
   for (size_t j = 0; j < committed_idx.size(); ++j) {
     PutCompleteRequest pc{items[committed_idx[j]].key, digests[j]};
     auto it = shard_digests.find(committed_idx[j]);
     if (it != shard_digests.end()) pc.shard_digests = std::move(it->second);
     completes.reqs.push_back(std::move(pc));
+    complete_order.push_back(committed_idx[j]);
   }
-  for (size_t j = 0; j < fused_hash_idx.size(); ++j)
+  for (size_t j = 0; j < fused_hash_idx.size(); ++j) {
     completes.reqs.push_back(
         PutCompleteRequest{items[fused_hash_idx[j]].key, fused_digests[j]});
+    complete_order.push_back(fused_hash_idx[j]);
+  }
   if (!completes.reqs.empty()) {
     auto r = c_.meta_call<PutCompleteListMsg, StatusListMsg>(
         M::BATCH_PUT_COMPLETE, completes);
     if (!r.ok()) return r.error();
+    for (size_t j = 0; j < complete_order.size() && j < r->statuses.size(); ++j)
+      if (r->statuses[j] != 0) statuses[complete_order[j]] = r->statuses[j];
   }
   std::vector<std::string> cancels;
   for (size_t i = 0; i < items.size(); ++i)
@@ -1083,6 +1112,26 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device(
 }
 
 Result<std::vector<int32_t>> GpuClient::batch_get_device(
+    const std::vector<DevGetItem>& items, bool verify, BatchGetSession* sess) {
+  const uint64_t gen = c_.reconnect_generation();
+  auto st = batch_get_device_once(items, verify, sess);
+  if (!st.ok() || c_.reconnect_generation() == gen) return st;
+  std::vector<DevGetItem> redo;
+  std::vector<size_t> redo_idx;
+  for (size_t i = 0; i < items.size(); ++i)
+    if (Client::failover_retriable(st.value()[i])) {
+      redo.push_back(items[i]);
+      redo_idx.push_back(i);
+    }
+  if (redo.empty()) return st;
+  auto st2 = batch_get_device_once(redo, verify, nullptr);
+  if (!st2.ok()) return st;
+  for (size_t j = 0; j < redo_idx.size(); ++j)
+    st.value()[redo_idx[j]] = st2.value()[j];
+  return st;
+}
+
+Result<std::vector<int32_t>> GpuClient::batch_get_device_once(
     const std::vector<DevGetItem>& items, bool verify, BatchGetSession* sess) {
   BB_RETURN_IF_ERROR(init());
   BB_HIP(hipSetDevice(device_));

@@ -69,11 +69,19 @@ class Client {
     const void* data;
     uint64_t size;
   };
-  // One metadata RPC for the whole batch; per-item status out.
+  // One metadata RPC for the whole batch; per-item status out. A leader
+  // failover mid-batch (put_start answered by the old leader, put_complete
+  // by the new one that never saw the PENDING objects) is resumed
+  // transparently: items that failed with failover-shaped errors WHILE a
+  // reconnect happened are redone once against the new leader.
   Result<std::vector<int32_t>> batch_put(const std::vector<PutItem>& items,
                                          const PlacementConfig& cfg = {});
   Result<std::vector<std::pair<int32_t, std::string>>> batch_get(
       const std::vector<ObjectKey>& keys);
+  // bumped every time meta_call_raw re-established the leader connection
+  uint64_t reconnect_generation() const { return reconnect_gen_.load(); }
+  // status looks like a lost-leader symptom (worth one redo after failover)?
+  static bool failover_retriable(int32_t st);
 
   // ------------------------------------------------------ cluster view
   Result<ClusterStats> cluster_stats();
@@ -125,6 +133,13 @@ class Client {
   std::mutex pool_cache_mu_;
   std::map<PoolId, AccessInfo> pool_cache_;
   std::mutex reconnect_mu_;  // one thread rediscovers/reconnects at a time
+  std::atomic<uint64_t> reconnect_gen_{0};
+  Result<std::vector<int32_t>> batch_put_once(const std::vector<PutItem>& items,
+                                              const PlacementConfig& cfg);
+  Result<void> put_once(const ObjectKey& key, const void* data, uint64_t size,
+                        const PlacementConfig& cfg);
+  Result<std::vector<std::pair<int32_t, std::string>>> batch_get_once(
+      const std::vector<ObjectKey>& keys);
 };
 
 }  // namespace blackbird

@@ -136,6 +136,12 @@ class GpuClient {
   // Resolve a shard to a device-visible pointer (local or IPC-mapped peer
   // HBM); .ptr nullptr if the pool is not device-visible from this process.
   Resolved resolve_device_ptr(const ShardPlacement& s);
+  Result<std::vector<int32_t>> batch_put_device_once(
+      const std::vector<DevPutItem>& items, const PlacementConfig& cfg,
+      BatchPutSession* sess);
+  Result<std::vector<int32_t>> batch_get_device_once(
+      const std::vector<DevGetItem>& items, bool verify,
+      BatchGetSession* sess);
   // device-visible base of a pool for the v2 batch protocols (HBM local,
   // IPC peer, or GPU-mapped host tier); nullptr → staged fallback
   uint8_t* resolve_pool_base(const PoolId& pool_id, AccessInfo* access,

@@ -156,6 +156,10 @@ void KeystoneServer::register_handlers() {
     return serde::to_bytes(U64Msg{moved.value()});
   });
   rpc_.register_handler(M::BATCH_PUT_START, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
+    // standby answers with a CALL-level error (not per-item statuses) so the
+    // client's failover machinery rediscovers the leader and retries
+    if (!ks.is_leader())
+      return Error{ErrorCode::NOT_LEADER, "standby keystone; retry on leader"};
     auto r = decode<BatchPutStartRequest>(b);
     if (!r.ok()) return r.error();
     return serde::to_bytes(ks.batch_put_start(r->requests));
@@ -177,6 +181,8 @@ void KeystoneServer::register_handlers() {
   });
   // ---- compact v2 batch protocol: pool-table + fixed-width placements ----
   rpc_.register_handler(M::BATCH_PUT_START2, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
+    if (!ks.is_leader())
+      return Error{ErrorCode::NOT_LEADER, "standby keystone; retry on leader"};
     serde::Dec d(b.data(), b.size());
     uint32_t count = d.num<uint32_t>();
     uint64_t uniform = d.num<uint64_t>();

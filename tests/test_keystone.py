@@ -510,3 +510,99 @@ class TestPutSessions:
         ks.put_complete("shape", checksum=9)
         with pytest.raises(Exception, match="SESSION_STALE"):
             ks.upsert_start_token(tok)
+
+
+class TestFailoverMidBatch:
+    def test_batches_survive_leader_failover_with_zero_errors(self):
+        """a keystone leader failover in the MIDDLE of a batch
+        workload must be absorbed by the client — every batch_put/batch_get
+        reports success; items whose put_start the old leader answered are
+        redone against the new leader transparently."""
+        import os as _os
+        import threading
+        cs = bb.CoordServer()
+        cs.start("127.0.0.1", 0)
+        ep = "127.0.0.1:%d" % cs.port
+
+        def mk_ks():
+            cfg = bb.KeystoneConfig()
+            cfg.listen_address = "127.0.0.1:0"
+            cfg.coord_endpoint = ep
+            cfg.enable_ha = True
+            cfg.persist_objects = True
+            cfg.worker_ttl_ms = 600
+            cfg.gc_interval_ms = 100000
+            return bb.create_and_start_keystone(cfg)
+
+        k1 = mk_ks()
+        deadline = time.time() + 5
+        while time.time() < deadline and not k1.service().is_leader():
+            time.sleep(0.02)
+        assert k1.service().is_leader()
+        k2 = mk_ks()
+
+        wc = bb.WorkerConfig()
+        wc.worker_id = "mbw0"
+        wc.coord_endpoint = ep
+        wc.data_listen_address = "127.0.0.1:0"
+        wc.heartbeat_interval_ms = 200
+        wc.heartbeat_ttl_ms = 5000
+        p = bb.PoolConfig()
+        p.pool_id = "mbpool"
+        p.storage_class = bb.StorageClass.RAM_CPU
+        p.size_bytes = 128 << 20
+        wc.pools = [p]
+        w = bb.WorkerService(wc)
+        w.initialize()
+        w.start()
+        deadline = time.time() + 5
+        while (time.time() < deadline and
+               not k1.service().get_memory_pools()):
+            time.sleep(0.02)
+
+        o = bb.ClientOptions()
+        o.keystone_endpoint = ""
+        o.coord_endpoint = ep
+        c = bb.Client(o)
+        c.connect()
+
+        stop_at = time.time() + 1.0
+        killed = threading.Event()
+
+        def killer():
+            while time.time() < stop_at:
+                time.sleep(0.02)
+            k1.stop()
+            k1.service().stop()  # releases the lease; k2 campaigns
+            killed.set()
+
+        kt = threading.Thread(target=killer)
+        kt.start()
+        try:
+            B = 32
+            blobs = [_os.urandom(32 * 1024) for _ in range(B)]
+            step = 0
+            end = time.time() + 4.0
+            while time.time() < end or not killed.is_set():
+                keys = ["mb-%d-%d" % (step, i) for i in range(B)]
+                st = c.batch_put(list(zip(keys, blobs)))
+                assert all(s == 0 for s in st), (step, st[:5])
+                res = c.batch_get(keys)
+                assert all(s == 0 for s, _ in res), (step,
+                                                     [s for s, _ in res][:5])
+                for i, (s, v) in enumerate(res):
+                    assert v == blobs[i], (step, i)
+                st = c.batch_remove(keys)
+                assert all(s == 0 for s in st), (step, st[:5])
+                step += 1
+                if time.time() > end + 20:
+                    raise TimeoutError("failover never absorbed")
+            assert step >= 2  # batches ran before AND after the kill
+            assert k2.service().is_leader()
+        finally:
+            kt.join()
+            c.close()
+            w.stop()
+            k2.stop()
+            k2.service().stop()
+            cs.stop()
