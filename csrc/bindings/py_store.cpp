@@ -481,6 +481,8 @@ void bind_store(py::module_& m) {
            }, py::arg("batch"), py::arg("verify") = false)
       .def_property_readonly("session_put_steps", &GpuClient::session_put_steps)
       .def_property_readonly("session_get_steps", &GpuClient::session_get_steps)
+      .def_property_readonly("session_graph_steps",
+                             &GpuClient::session_graph_steps)
       .def("batch_put_async",
            [](GpuClient& g,
               const std::vector<std::tuple<std::string, uint64_t, uint64_t>>& items,

@@ -2,6 +2,7 @@
 
 #include <algorithm>
 #include <atomic>
+#include <cstdlib>
 #include <cstring>
 #include <future>
 #include <map>
@@ -334,6 +335,30 @@ struct PoolRef {
 };
 }  // namespace
 
+template <typename Sess>
+Result<void> GpuClient::session_kernel(Sess* sess, uint64_t* digests) {
+  static const bool no_graph = std::getenv("BB_NO_HIPGRAPH") != nullptr;
+  if (!no_graph && !sess->plan && !sess->plan_failed) {
+    auto plan = std::make_shared<gpu::FusedPutPlan>();
+    auto rb = plan->build(sess->descs.data(),
+                          static_cast<uint32_t>(sess->descs.size()), device_);
+    if (rb.ok()) sess->plan = std::move(plan);
+    else sess->plan_failed = true;  // capture unsupported: launch path
+  }
+  if (sess->plan) {
+    auto r = sess->plan->run(digests);
+    if (r.ok()) {
+      session_graph_steps_.fetch_add(1);
+      return r;
+    }
+    sess->plan.reset();  // replay failed: fall through to launches
+    sess->plan_failed = true;
+  }
+  return gpu::fused_put(sess->descs.data(),
+                        static_cast<uint32_t>(sess->descs.size()), digests,
+                        streams_[2]);
+}
+
 // Token fast path: placements unchanged since last step ⇒ two tiny RPCs
 // bracket ONE fused copy+digest kernel launch. Returns nullopt when the
 // session is not usable (caller runs the full path, which re-establishes it).
@@ -368,9 +393,7 @@ std::optional<Result<std::vector<int32_t>>> GpuClient::try_session_put(
     return std::nullopt;
   }
   std::vector<uint64_t> digests(items.size(), 0);
-  auto rk = gpu::fused_put(sess->descs.data(),
-                           static_cast<uint32_t>(sess->descs.size()),
-                           digests.data(), streams_[2]);
+  auto rk = session_kernel(sess, digests.data());
   if (!rk.ok()) return {rk.error()};  // objects stay PENDING; GC reclaims
   serde::Enc e2;
   e2.num<uint64_t>(sess->token);
@@ -571,6 +594,8 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
     if (sess && token != 0 && fused_hash_idx.size() == items.size()) {
       bool all_cached = true;
       sess->descs = fused_hash;
+      sess->plan.reset();  // descs changed: a kept plan would replay stale
+      sess->plan_failed = false;
       sess->entries.clear();
       sess->entries.reserve(items.size());
       for (size_t j = 0; j < items.size() && all_cached; ++j) {
@@ -648,9 +673,7 @@ std::optional<Result<std::vector<int32_t>>> GpuClient::try_session_get(
     }
   BB_TRACE_SCOPE("bb::session_get");
   std::vector<uint64_t> got(items.size(), 0);
-  auto rk = gpu::fused_put(sess->descs.data(),
-                           static_cast<uint32_t>(sess->descs.size()),
-                           got.data(), streams_[2]);
+  auto rk = session_kernel(sess, got.data());
   if (!rk.ok()) return {rk.error()};
   std::vector<uint32_t> miss;
   for (size_t j = 0; j < items.size(); ++j)
@@ -750,6 +773,8 @@ Result<std::vector<int32_t>> GpuClient::batch_get_device_v2(
     if (sess && all_verified && miss_idx.empty() &&
         hit_idx.size() == items.size() && epoch_at_lookup == cache_epoch_) {
       sess->descs = std::move(descs);
+      sess->plan.reset();
+      sess->plan_failed = false;
       sess->entries.assign(hit_entries.begin(), hit_entries.end());
       sess->cache_epoch = cache_epoch_;
       sess->owner = this;

@@ -10,6 +10,7 @@
 #include <hip/hip_runtime.h>
 #include <cstring>
 
+#include <memory>
 #include <vector>
 
 #include "blackbird/gpu/digest_spec.h"
@@ -380,6 +381,104 @@ Result<void> fused_put(const PutDesc* descs, uint32_t n, uint64_t* out_digests,
   return {};
 }
 
+// ----------------- hipGraph-replayed session step (FusedPutPlan) -----------
+// A batch session re-runs the SAME desc list every step, so the descriptors
+// are uploaded once and the step is captured as a graph: one hipGraphLaunch
+// replaces memset + 2 kernel launches + D2H issue each step.
 
+struct FusedPutPlan::Impl {
+  hipStream_t stream = nullptr;   // dedicated capture/replay stream
+  hipGraphExec_t exec = nullptr;
+  void* dev = nullptr;            // segs + prefix + out
+  uint64_t* h_out = nullptr;      // pinned digest landing zone
+  size_t out_bytes = 0;
+
+  ~Impl() {
+    if (exec) (void)hipGraphExecDestroy(exec);
+    if (dev) (void)hipFree(dev);
+    if (h_out) (void)hipHostFree(h_out);
+    if (stream) (void)hipStreamDestroy(stream);
+  }
+};
+
+FusedPutPlan::~FusedPutPlan() { delete impl_; }
+
+Result<void> FusedPutPlan::build(const PutDesc* descs, uint32_t n, int device) {
+  if (impl_) return Error{ErrorCode::INVALID_ARGUMENT, "plan already built"};
+  if (n == 0) return Error{ErrorCode::INVALID_ARGUMENT, "empty desc list"};
+  BB_HIP_TRY(hipSetDevice(device));
+  auto impl = std::make_unique<Impl>();
+  const size_t segs_bytes = n * sizeof(PutSeg);
+  const size_t prefix_bytes = n * sizeof(uint64_t);
+  impl->out_bytes = n * sizeof(uint64_t);
+
+  std::vector<PutSeg> h_segs(n);
+  std::vector<uint64_t> h_prefix(n);
+  uint64_t total = 0;
+  for (uint32_t i = 0; i < n; ++i) {
+    h_segs[i] = {static_cast<const uint8_t*>(descs[i].src),
+                 static_cast<uint8_t*>(descs[i].dst), descs[i].nbytes};
+    h_prefix[i] = total;
+    total += (descs[i].nbytes + blackbird::digest::kTileBytes - 1) /
+             blackbird::digest::kTileBytes;
+  }
+  if (total == 0) return Error{ErrorCode::INVALID_ARGUMENT, "zero-byte batch"};
+
+  BB_HIP_TRY(hipStreamCreateWithFlags(&impl->stream, hipStreamNonBlocking));
+  BB_HIP_TRY(hipMalloc(&impl->dev, segs_bytes + prefix_bytes + impl->out_bytes));
+  BB_HIP_TRY(hipHostMalloc(reinterpret_cast<void**>(&impl->h_out),
+                           impl->out_bytes, hipHostMallocDefault));
+  auto* d_segs = static_cast<PutSeg*>(impl->dev);
+  auto* d_prefix = reinterpret_cast<uint64_t*>(
+      static_cast<uint8_t*>(impl->dev) + segs_bytes);
+  auto* d_out = reinterpret_cast<unsigned long long*>(
+      static_cast<uint8_t*>(impl->dev) + segs_bytes + prefix_bytes);
+  BB_HIP_TRY(hipMemcpy(d_segs, h_segs.data(), segs_bytes,
+                       hipMemcpyHostToDevice));
+  BB_HIP_TRY(hipMemcpy(d_prefix, h_prefix.data(), prefix_bytes,
+                       hipMemcpyHostToDevice));
+
+  uint64_t waves = (total + 1) / 2;
+  uint64_t blocks = std::min<uint64_t>((waves + 3) / 4, 4096);
+  if (blocks < 1) blocks = 1;
+
+  hipGraph_t graph = nullptr;
+  BB_HIP_TRY(hipStreamBeginCapture(impl->stream,
+                                   hipStreamCaptureModeThreadLocal));
+  hipError_t cap = hipSuccess;
+  do {
+    if ((cap = hipMemsetAsync(d_out, 0, impl->out_bytes, impl->stream)) !=
+        hipSuccess)
+      break;
+    fused_put_kernel<<<static_cast<int>(blocks), kBlock, 0, impl->stream>>>(
+        d_segs, d_prefix, n, total, d_out);
+    if ((cap = hipGetLastError()) != hipSuccess) break;
+    fused_put_finalize_kernel<<<(n + 255) / 256, 256, 0, impl->stream>>>(
+        d_segs, n, d_out);
+    if ((cap = hipGetLastError()) != hipSuccess) break;
+    cap = hipMemcpyAsync(impl->h_out, d_out, impl->out_bytes,
+                         hipMemcpyDeviceToHost, impl->stream);
+  } while (false);
+  hipError_t endc = hipStreamEndCapture(impl->stream, &graph);
+  if (cap != hipSuccess) {
+    if (graph) (void)hipGraphDestroy(graph);
+    return hip_error(cap, "graph capture");
+  }
+  BB_HIP_TRY(endc);
+  hipError_t inst = hipGraphInstantiate(&impl->exec, graph, nullptr, nullptr, 0);
+  (void)hipGraphDestroy(graph);
+  BB_HIP_TRY(inst);
+  n_ = n;
+  impl_ = impl.release();
+  return {};
+}
+
+Result<void> FusedPutPlan::run(uint64_t* out_digests) {
+  if (!impl_) return Error{ErrorCode::INVALID_ARGUMENT, "plan not built"};
+  BB_HIP_TRY(hipGraphLaunch(impl_->exec, impl_->stream));
+  BB_HIP_TRY(hipStreamSynchronize(impl_->stream));
+  std::memcpy(out_digests, impl_->h_out, impl_->out_bytes);
+  return {};
+}
 
 }  // namespace blackbird::gpu

@@ -14,6 +14,7 @@
 #pragma once
 
 #include <future>
+#include <memory>
 #include <mutex>
 #include <optional>
 #include <unordered_map>
@@ -67,6 +68,10 @@ class GpuClient {
                             // that client's placement cache)
     std::vector<gpu::PutDesc> descs;  // src=user buffer, dst=pool range
     std::vector<void*> entries;       // CachedPlacement* digest slots
+    // hipGraph replay of the step (descs fixed ⇒ captured once, one graph
+    // launch per step); built lazily on the first session step
+    std::shared_ptr<gpu::FusedPutPlan> plan;
+    bool plan_failed = false;  // capture unsupported: stay on fused_put
   };
   struct BatchGetSession {
     bool complete = false;  // covers every item
@@ -74,6 +79,8 @@ class GpuClient {
     void* owner = nullptr;
     std::vector<gpu::PutDesc> descs;  // src=pool range, dst=user buffer
     std::vector<void*> entries;       // CachedPlacement* want-digest slots
+    std::shared_ptr<gpu::FusedPutPlan> plan;
+    bool plan_failed = false;
   };
 
   // One metadata RPC + fused transfers + one batched checksum launch.
@@ -86,6 +93,8 @@ class GpuClient {
   // fast-path step counters (tests/bench introspection)
   uint64_t session_put_steps() const { return session_put_steps_; }
   uint64_t session_get_steps() const { return session_get_steps_; }
+  // session steps that replayed the captured hipGraph (vs per-op launches)
+  uint64_t session_graph_steps() const { return session_graph_steps_; }
 
   // ---- pipelined batches: begin returns a token immediately; the batch
   // runs on a background thread (metadata RPCs of batch N+1 overlap the
@@ -201,6 +210,12 @@ class GpuClient {
   uint64_t cache_epoch_ = 0;
   std::atomic<uint64_t> session_put_steps_{0};
   std::atomic<uint64_t> session_get_steps_{0};
+  std::atomic<uint64_t> session_graph_steps_{0};
+  // run the session desc list: captured-graph replay when available,
+  // fused_put launches otherwise (builds the plan lazily; BB_NO_HIPGRAPH=1
+  // or a failed capture pins the session to the launch path)
+  template <typename Sess>
+  Result<void> session_kernel(Sess* sess, uint64_t* digests);
   bool fused_copy_ = true;
   bool initialized_ = false;
 };

@@ -81,6 +81,30 @@ struct PutDesc {
 Result<void> fused_put(const PutDesc* descs, uint32_t n, uint64_t* out_digests,
                        hipStream_t stream);
 
+// Pre-instantiated fused-put step for batch sessions (fixed desc list every
+// step): descriptors live on-device (uploaded once by build), and the whole
+// step — digest reset → fused copy+digest kernel → finalize → digest D2H —
+// is captured in a hipGraph and replayed as ONE graph launch per step
+// instead of 5 stream ops. run() blocks until the digests are in
+// out_digests[n].
+class FusedPutPlan {
+ public:
+  FusedPutPlan() = default;
+  ~FusedPutPlan();
+  FusedPutPlan(const FusedPutPlan&) = delete;
+  FusedPutPlan& operator=(const FusedPutPlan&) = delete;
+
+  Result<void> build(const PutDesc* descs, uint32_t n, int device);
+  bool ready() const { return impl_ != nullptr; }
+  uint32_t size() const { return n_; }
+  Result<void> run(uint64_t* out_digests);
+
+ private:
+  struct Impl;
+  Impl* impl_ = nullptr;
+  uint32_t n_ = 0;
+};
+
 // Simple utilities used by tests/benchmarks.
 Result<void> fill_pattern(void* dev_ptr, uint64_t nbytes, uint64_t seed,
                           hipStream_t stream);

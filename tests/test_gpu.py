@@ -672,6 +672,9 @@ class TestBatchSessions:
             assert gcl.session_put_steps >= 3, gcl.session_put_steps
             assert gcl.session_get_steps >= 2, gcl.session_get_steps
             assert ks.token_commits() >= 3
+            # session steps replay the captured hipGraph, not per-op launches
+            if not os.environ.get("BB_NO_HIPGRAPH"):
+                assert gcl.session_graph_steps >= 3, gcl.session_graph_steps
 
             # server-side interference: removing ANY object bumps the
             # placement epoch → session falls back transparently, still OK
