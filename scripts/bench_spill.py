@@ -70,13 +70,16 @@ all_keys = []
 t0 = time.perf_counter()
 bytes_moved = 0
 get_lat = []
+phases = os.environ.get("BB_SPILL_PHASES") == "1"
 for step in range(STEPS):
     keys = ["sp-%d-%d" % (step, i) for i in range(BATCH)]
+    tp = time.perf_counter()
     if use_gpu:
         st = gcl.batch_put_device([(k, src + i * OBJ, OBJ)
                                    for i, k in enumerate(keys)], cfg)
     else:
         st = client.batch_put([(k, b"\xab" * OBJ) for k in keys], cfg)
+    put_ms = (time.perf_counter() - tp) * 1e3
     placed = [k for k, s in zip(keys, st) if s == 0]
     all_keys.extend(placed)
     bytes_moved += len(placed) * OBJ
@@ -92,6 +95,9 @@ for step in range(STEPS):
     ok = sum(1 for s in st if s == 0)
     assert ok == len(sample), st[:5]
     bytes_moved += ok * OBJ
+    if phases:
+        print("[phase] step=%d put_ms=%.1f get_ms=%.1f" %
+              (step, put_ms, get_lat[-1]), file=sys.stderr, flush=True)
 elapsed = time.perf_counter() - t0
 
 dist = {}
