@@ -81,8 +81,8 @@ def max_over_ranks(dist, value):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=10)
-    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--steps", type=int, default=50)
+    ap.add_argument("--warmup", type=int, default=10)
     ap.add_argument("--object-size", type=int, default=1 * MB)
     ap.add_argument("--objects", type=int, default=512,
                     help="objects per rank per step")

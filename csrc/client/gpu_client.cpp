@@ -63,7 +63,29 @@ GpuClient::~GpuClient() {
     for (auto& s : streams_)
       if (s) (void)hipStreamDestroy(s);
     if (staging_) (void)hipHostFree(staging_);
+    for (void* p : staging_pool_) (void)hipHostFree(p);
   }
+}
+
+void* GpuClient::acquire_staging_buf() {
+  {
+    std::lock_guard<std::mutex> g(staging_pool_mu_);
+    if (!staging_pool_.empty()) {
+      void* p = staging_pool_.back();
+      staging_pool_.pop_back();
+      return p;
+    }
+  }
+  void* p = nullptr;
+  if (hipHostMalloc(&p, kFanBuf, hipHostMallocDefault) != hipSuccess)
+    return nullptr;
+  return p;
+}
+
+void GpuClient::release_staging_buf(void* p) {
+  if (!p) return;
+  std::lock_guard<std::mutex> g(staging_pool_mu_);
+  staging_pool_.push_back(p);
 }
 
 Result<void> GpuClient::init() {
@@ -197,25 +219,24 @@ Result<void> GpuClient::staged_write_many(
     const std::vector<std::pair<ShardPlacement, const void*>>& work) {
   if (work.empty()) return {};
   if (work.size() == 1) return staged_write(work[0].first, work[0].second);
-  const int nthreads = std::min<int>(4, static_cast<int>(work.size()));
+  const int nthreads = std::min<int>(kFanThreads,
+                                     static_cast<int>(work.size()));
   std::atomic<size_t> next{0};
   std::vector<std::future<Result<void>>> futs;
-  const uint64_t per_buf = 16ull << 20;
   for (int t = 0; t < nthreads; ++t)
     futs.push_back(std::async(std::launch::async, [&]() -> Result<void> {
-      void* buf = nullptr;
-      if (hipHostMalloc(&buf, per_buf, hipHostMallocDefault) != hipSuccess)
-        return Error{ErrorCode::HIP_ERROR, "staging alloc"};
+      void* buf = acquire_staging_buf();
+      if (!buf) return Error{ErrorCode::HIP_ERROR, "staging alloc"};
       Result<void> rc{};
       for (size_t i = next.fetch_add(1); i < work.size();
            i = next.fetch_add(1)) {
-        auto r = staged_write_buf(work[i].first, work[i].second, buf, per_buf);
+        auto r = staged_write_buf(work[i].first, work[i].second, buf, kFanBuf);
         if (!r.ok()) {
           rc = r;
           break;
         }
       }
-      (void)hipHostFree(buf);
+      release_staging_buf(buf);
       return rc;
     }));
   for (auto& f : futs) BB_RETURN_IF_ERROR(f.get());
@@ -231,25 +252,24 @@ Result<void> GpuClient::staged_read_many(
     const std::vector<std::pair<ShardPlacement, void*>>& work) {
   if (work.empty()) return {};
   if (work.size() == 1) return staged_read(work[0].first, work[0].second);
-  const int nthreads = std::min<int>(4, static_cast<int>(work.size()));
+  const int nthreads = std::min<int>(kFanThreads,
+                                     static_cast<int>(work.size()));
   std::atomic<size_t> next{0};
   std::vector<std::future<Result<void>>> futs;
-  const uint64_t per_buf = 16ull << 20;
   for (int t = 0; t < nthreads; ++t)
     futs.push_back(std::async(std::launch::async, [&]() -> Result<void> {
-      void* buf = nullptr;
-      if (hipHostMalloc(&buf, per_buf, hipHostMallocDefault) != hipSuccess)
-        return Error{ErrorCode::HIP_ERROR, "staging alloc"};
+      void* buf = acquire_staging_buf();
+      if (!buf) return Error{ErrorCode::HIP_ERROR, "staging alloc"};
       Result<void> rc{};
       for (size_t i = next.fetch_add(1); i < work.size();
            i = next.fetch_add(1)) {
-        auto r = staged_read_buf(work[i].first, work[i].second, buf, per_buf);
+        auto r = staged_read_buf(work[i].first, work[i].second, buf, kFanBuf);
         if (!r.ok()) {
           rc = r;
           break;
         }
       }
-      (void)hipHostFree(buf);
+      release_staging_buf(buf);
       return rc;
     }));
   for (auto& f : futs) BB_RETURN_IF_ERROR(f.get());

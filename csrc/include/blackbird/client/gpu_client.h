@@ -183,6 +183,10 @@ class GpuClient {
   // fan staged reads (host/TCP pools) out over a small pinned-buffer pool
   Result<void> staged_read_many(
       const std::vector<std::pair<ShardPlacement, void*>>& work);
+  // reusable pinned staging buffers for the fan-out paths (hipHostMalloc is
+  // milliseconds per call — allocating per batch dominated the NVMe legs)
+  void* acquire_staging_buf();
+  void release_staging_buf(void* p);
 
   Client& c_;
   int device_;
@@ -195,6 +199,10 @@ class GpuClient {
   void* staging_ = nullptr;  // pinned bounce buffer for TCP/SHM pools
   uint64_t staging_size_ = 64ull << 20;
   std::mutex staging_mu_;  // async batches share the bounce buffer
+  static constexpr uint64_t kFanBuf = 16ull << 20;  // per-thread fan-out buf
+  static constexpr int kFanThreads = 8;
+  std::mutex staging_pool_mu_;
+  std::vector<void*> staging_pool_;  // idle pinned kFanBuf buffers
   struct CachedPlacement {
     PoolId pool_id;
     uint64_t offset = 0;
