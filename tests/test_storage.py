@@ -99,6 +99,27 @@ class TestDiskPersistence:
         assert len(files) == 1
         assert files[0].stat().st_size == 1 * MB
 
+    def test_direct_io_length_matrix(self, tmp_path):
+        """Round-trip every shape class through the direct-IO pool: 4 KiB
+        multiples ride the zero-copy pread/pwrite fast path (large Python
+        bytes are page-aligned), odd lengths exercise the bounced tail, and
+        >8 MiB ops split across the IO thread pool."""
+        b = make(bb.StorageClass.NVME, size=64 * MB, tmpdir=tmp_path,
+                 pool_id="matrix")
+        import hashlib
+        for length in [4096, 1 * MB, 1 * MB + 100, 12345,
+                       8 * MB + 4096, 9 * MB + 7]:
+            t = b.reserve(length)
+            b.commit(t.token_id)
+            data = os.urandom(length)
+            b.write(t.offset, data)
+            back = b.read(t.offset, length)
+            assert hashlib.sha256(back).digest() == \
+                hashlib.sha256(data).digest(), length
+            assert b.checksum(t.offset, length) == \
+                bb.core.gpu.checksum_cpu(data), length
+            b.free(t.offset, length)
+
     def test_invalid_dir_fails_loudly(self):
         cfg = bb.PoolConfig()
         cfg.pool_id = "bad"
