@@ -327,9 +327,16 @@ def main():
         cold_keys = [k for k, _, _ in ck]
 
         def cold_step():
+            t0 = time.perf_counter()
             assert gcl_cold.batch_put_prepared(cpb, cold_cfg), "cold put"
+            t1 = time.perf_counter()
             assert gcl_cold.batch_get_prepared(cgb), "cold get"
+            t2 = time.perf_counter()
             assert bb.client_batch_remove_prepared(lane_clients[0], cpb)
+            if phase_log:
+                t3 = time.perf_counter()
+                log(f"cold phases put={1e3*(t1-t0):.2f} get={1e3*(t2-t1):.2f} "
+                    f"rm={1e3*(t3-t2):.2f}")
 
         cold_step()  # warmup
         device_sync()

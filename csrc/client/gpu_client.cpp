@@ -417,6 +417,7 @@ std::optional<Result<std::vector<int32_t>>> GpuClient::try_session_put(
   if (!rk.ok()) return {rk.error()};  // objects stay PENDING; GC reclaims
   serde::Enc e2;
   e2.num<uint64_t>(sess->token);
+  e2.num<uint8_t>(0);  // keep the session for the next step
   e2.num<uint32_t>(static_cast<uint32_t>(digests.size()));
   for (uint64_t dg : digests) e2.num<uint64_t>(dg);
   auto r2 = c_.meta_call_raw(M::BATCH_COMMIT_TOKEN, e2.buf);
@@ -444,9 +445,14 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
     BatchPutSession* sess) {
   if (auto fast = try_session_put(items, sess)) return std::move(*fast);
   BB_TRACE_SCOPE("bb::batch_put");
-  const bool want_token = sess != nullptr && cfg.replace &&
-                          cfg.replication <= 1 && cfg.checksum &&
-                          placement_cache_on_ && fused_copy_;
+  // establish: a reusable session for upsert steps (token kept server-side)
+  const bool establish = sess != nullptr && cfg.replace &&
+                         cfg.replication <= 1 && cfg.checksum &&
+                         placement_cache_on_ && fused_copy_;
+  // one-shot tokens let ANY all-fused batch commit by token+digests instead
+  // of re-sending every key in BATCH_PUT_COMPLETE (released at commit)
+  const bool want_token = establish ||
+                          (cfg.replication <= 1 && cfg.checksum && fused_copy_);
   serde::Enc req;
   req.num<uint32_t>(static_cast<uint32_t>(items.size()));
   // uniform size when possible (the common batched pattern)
@@ -580,19 +586,34 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
         statuses[i] = static_cast<int32_t>(ErrorCode::TRANSFER_FAILED);
   }
 
-  PutCompleteListMsg completes;
-  for (size_t j = 0; j < committed_idx.size(); ++j) {
-    if (statuses[committed_idx[j]] != 0) continue;  // staged write failed
-    completes.reqs.push_back(
-        PutCompleteRequest{items[committed_idx[j]].key, digests[j]});
+  // commit: token path (8 B + digests, no key strings) when the whole batch
+  // rode the fused single-copy path; key-based BATCH_PUT_COMPLETE otherwise
+  // (or if the token went stale — concurrent placement change)
+  bool committed_by_token = false;
+  if (token != 0 && fused_hash_idx.size() == items.size()) {
+    serde::Enc e2;
+    e2.num<uint64_t>(token);
+    e2.num<uint8_t>(establish ? 0 : 1);  // one-shot: release after commit
+    e2.num<uint32_t>(static_cast<uint32_t>(fused_digests.size()));
+    for (uint64_t dg : fused_digests) e2.num<uint64_t>(dg);
+    auto r2 = c_.meta_call_raw(M::BATCH_COMMIT_TOKEN, e2.buf);
+    committed_by_token = r2.ok();
   }
-  for (size_t j = 0; j < fused_hash_idx.size(); ++j)
-    completes.reqs.push_back(
-        PutCompleteRequest{items[fused_hash_idx[j]].key, fused_digests[j]});
-  if (!completes.reqs.empty()) {
-    auto r = c_.meta_call<PutCompleteListMsg, StatusListMsg>(
-        M::BATCH_PUT_COMPLETE, completes);
-    if (!r.ok()) return r.error();
+  if (!committed_by_token) {
+    PutCompleteListMsg completes;
+    for (size_t j = 0; j < committed_idx.size(); ++j) {
+      if (statuses[committed_idx[j]] != 0) continue;  // staged write failed
+      completes.reqs.push_back(
+          PutCompleteRequest{items[committed_idx[j]].key, digests[j]});
+    }
+    for (size_t j = 0; j < fused_hash_idx.size(); ++j)
+      completes.reqs.push_back(
+          PutCompleteRequest{items[fused_hash_idx[j]].key, fused_digests[j]});
+    if (!completes.reqs.empty()) {
+      auto r = c_.meta_call<PutCompleteListMsg, StatusListMsg>(
+          M::BATCH_PUT_COMPLETE, completes);
+      if (!r.ok()) return r.error();
+    }
   }
   if (placement_cache_on_ && !fused_hash_idx.empty()) {
     // remember our own single-copy placements + digests for RPC-free
@@ -611,7 +632,8 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
     }
     // establish the batch session when the server granted a token AND every
     // item rode the fused single-copy path (so desc j == item j)
-    if (sess && token != 0 && fused_hash_idx.size() == items.size()) {
+    if (sess && establish && token != 0 &&
+        fused_hash_idx.size() == items.size()) {
       bool all_cached = true;
       sess->descs = fused_hash;
       sess->plan.reset();  // descs changed: a kept plan would replay stale

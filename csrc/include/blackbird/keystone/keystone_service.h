@@ -76,7 +76,10 @@ class KeystoneService {
   // loop at keystone_service.cpp:302-360.
   uint64_t create_put_session(const std::vector<PutStartRequest>& reqs);
   Result<void> upsert_start_token(uint64_t token);
-  Result<void> commit_token(uint64_t token, const std::vector<uint64_t>& digests);
+  // release=true erases the session after a successful commit (one-shot
+  // token commits from cold batch puts — no key strings on the wire)
+  Result<void> commit_token(uint64_t token, const std::vector<uint64_t>& digests,
+                            bool release = false);
   uint64_t token_commits() const { return ctr_token_commits_.load(); }
 
   // ------------------------------------------------------ cluster view

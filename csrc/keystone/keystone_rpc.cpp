@@ -289,12 +289,13 @@ void KeystoneServer::register_handlers() {
   rpc_.register_handler(M::BATCH_COMMIT_TOKEN, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
     serde::Dec d(b.data(), b.size());
     uint64_t token = d.num<uint64_t>();
+    uint8_t flags = d.num<uint8_t>();  // bit 0: release session after commit
     uint32_t n = d.num<uint32_t>();
     if (!d.ok() || d.remaining() != n * sizeof(uint64_t) || n > (1u << 24))
       return Error{ErrorCode::PROTOCOL_ERROR, "bad commit request"};
     std::vector<uint64_t> digests(n);
     for (uint32_t i = 0; i < n; ++i) digests[i] = d.num<uint64_t>();
-    BB_RETURN_IF_ERROR(ks.commit_token(token, digests));
+    BB_RETURN_IF_ERROR(ks.commit_token(token, digests, flags & 1));
     return std::string{};
   });
   rpc_.register_handler(M::BATCH_REMOVE, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {

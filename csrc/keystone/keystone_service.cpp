@@ -490,7 +490,7 @@ Result<void> KeystoneService::upsert_start_token(uint64_t token) {
 }
 
 Result<void> KeystoneService::commit_token(
-    uint64_t token, const std::vector<uint64_t>& digests) {
+    uint64_t token, const std::vector<uint64_t>& digests, bool release) {
   if (!is_leader())
     return Error{ErrorCode::NOT_LEADER, "standby keystone; retry on leader"};
   std::shared_ptr<PutSession> s;
@@ -523,6 +523,11 @@ Result<void> KeystoneService::commit_token(
   }
   bump_view();
   ctr_token_commits_.fetch_add(1);
+  lk.unlock();
+  if (release) {
+    std::lock_guard<std::mutex> g(sessions_mu_);
+    put_sessions_.erase(token);
+  }
   return {};
 }
 
