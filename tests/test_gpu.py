@@ -719,6 +719,56 @@ class TestBatchSessions:
         finally:
             cl.stop()
 
+    def test_one_shot_token_commit(self):
+        """Even without a reusable session (cache off, replace off), an
+        all-fused batch put commits by token + digests — BATCH_PUT_COMPLETE
+        never re-sends the keys — and the one-shot token is released at
+        commit. Data and digests stay correct."""
+        cl = Cluster(n_workers=1, pool_bytes=256 * MB,
+                     storage_class=bb.StorageClass.RAM_GPU)
+        g = bb.core.gpu
+        try:
+            c = cl.client()
+            gcl = bb.GpuClient(c, 0)
+            gcl.init()  # placement cache OFF
+            N, S = 16, 32 * 1024
+            src = g.malloc(N * S)
+            dst = g.malloc(N * S)
+            blobs = [os.urandom(S) for _ in range(N)]
+            for i, b in enumerate(blobs):
+                g.upload(src + i * S, b)
+            cfg = bb.PlacementConfig()
+            cfg.replace = False
+            cfg.checksum = True
+            ks = cl.keystone.service()
+            before = ks.token_commits()
+            keys = ["ot%02d" % i for i in range(N)]
+            assert gcl.batch_put_device(
+                [(k, src + i * S, S) for i, k in enumerate(keys)],
+                cfg) == [0] * N
+            assert ks.token_commits() == before + 1
+            assert gcl.batch_get_device(
+                [(k, dst + i * S, S) for i, k in enumerate(keys)],
+                verify=True) == [0] * N
+            for i, b in enumerate(blobs):
+                assert g.download(dst + i * S, S) == b, i
+            info = ks.get_workers(keys[3])
+            assert info.checksum == g.checksum_cpu(blobs[3])
+            # replace=False still refuses overwrites after a token commit
+            st = gcl.batch_put_device([(keys[0], src, S)], cfg)
+            assert st[0] != 0
+            c.batch_remove(keys)
+            # fresh placements after remove: the one-shot path works again
+            assert gcl.batch_put_device(
+                [(k, src + i * S, S) for i, k in enumerate(keys)],
+                cfg) == [0] * N
+            assert ks.token_commits() == before + 2
+            g.free(src)
+            g.free(dst)
+            c.close()
+        finally:
+            cl.stop()
+
 
 class TestRcclShuffle:
     def test_two_rank_store_shuffle(self, tmp_path):
