@@ -506,9 +506,16 @@ Result<bool> Client::exists(const ObjectKey& key) {
 }
 
 Result<void> Client::remove(const ObjectKey& key) {
+  const uint64_t gen = reconnect_gen_.load();
   auto r = meta_call_raw(M::REMOVE_OBJECT, serde::to_bytes(KeyMsg{key}),
                           opts_.rpc_timeout_ms);
-  if (!r.ok()) return r.error();
+  if (!r.ok()) {
+    // see batch_remove: NOT_FOUND after an in-call failover = already gone
+    if (r.code() == ErrorCode::OBJECT_NOT_FOUND &&
+        reconnect_gen_.load() != gen)
+      return {};
+    return r.error();
+  }
   return {};
 }
 
@@ -524,9 +531,17 @@ Result<uint64_t> Client::remove_all() {
 
 Result<std::vector<int32_t>> Client::batch_remove(
     const std::vector<ObjectKey>& keys) {
+  const uint64_t gen = reconnect_gen_.load();
   auto r = meta_call<KeysMsg, StatusListMsg>(M::BATCH_REMOVE, KeysMsg{keys},
                                               opts_.rpc_timeout_ms);
   if (!r.ok()) return r.error();
+  if (reconnect_gen_.load() != gen) {
+    // a leader failover happened inside this call: the first attempt may
+    // have applied + replicated before the reply was lost, so the retried
+    // remove sees NOT_FOUND — the intent (key gone) is satisfied
+    for (auto& st : r->statuses)
+      if (st == static_cast<int32_t>(ErrorCode::OBJECT_NOT_FOUND)) st = 0;
+  }
   return std::move(r->statuses);
 }
 
@@ -571,8 +586,250 @@ Result<std::vector<int32_t>> Client::batch_put(const std::vector<PutItem>& items
   return st;
 }
 
+uint8_t* Client::host_pool_base(const PoolId& id, AccessInfo* access) {
+  bool is_dev = false;
+  if (void* b = LocalPools::inst().lookup(id, &is_dev)) {
+    // device pools stay on the shard path (write_shard/read_shard hipMemcpy)
+    return is_dev ? nullptr : static_cast<uint8_t*>(b);
+  }
+  auto a = pool_access(id);
+  if (!a.ok()) return nullptr;
+  uint8_t* out = nullptr;
+  if (a->kind == AccessKind::SHM && !a->shm_name.empty())
+    out = static_cast<uint8_t*>(mapper_->map_shm(a->shm_name, 0));
+  if (access) *access = std::move(a.value());
+  return out;
+}
+
+namespace {
+struct HostPoolRef {
+  PoolId pool_id;
+  uint8_t* base = nullptr;
+  AccessInfo access;
+};
+}  // namespace
+
+// Host twin of GpuClient::batch_put_device_v2: compact request, pool-table
+// response, memcpy into mapped pools (shard RPC fallback), digests computed
+// on the fly, commit by one-shot token when the whole batch qualifies —
+// BATCH_PUT_COMPLETE never re-sends keys on the common path.
+Result<std::vector<int32_t>> Client::batch_put_once_v2(
+    const std::vector<PutItem>& items, const PlacementConfig& cfg) {
+  serde::Enc req;
+  req.num<uint32_t>(static_cast<uint32_t>(items.size()));
+  uint64_t uniform = items.empty() ? 1 : items[0].size;
+  for (auto& it : items)
+    if (it.size != uniform) { uniform = 0; break; }
+  req.num<uint64_t>(uniform);
+  if (uniform == 0)
+    for (auto& it : items) req.num<uint64_t>(it.size);
+  for (auto& it : items) req.str(it.key);
+  serde::put(req, cfg);
+  const bool want_token = cfg.replication <= 1 && cfg.checksum;
+  req.num<uint8_t>(want_token ? 1 : 0);
+  auto resp = meta_call_raw(M::BATCH_PUT_START2, req.buf, opts_.rpc_timeout_ms);
+  if (!resp.ok()) return resp.error();
+
+  serde::Dec d(resp.value().data(), resp.value().size());
+  d.num<uint64_t>();  // view version
+  const uint64_t token = d.num<uint64_t>();
+  const uint16_t npools = d.num<uint16_t>();
+  std::vector<HostPoolRef> pools(npools);
+  for (uint16_t i = 0; i < npools; ++i) {
+    pools[i].pool_id = d.str();
+    pools[i].base = host_pool_base(pools[i].pool_id, &pools[i].access);
+  }
+
+  std::vector<int32_t> statuses(items.size(), 0);
+  // per-item placement list: (pool index, offset) per copy
+  std::vector<std::vector<std::pair<uint16_t, uint64_t>>> placed(items.size());
+  for (size_t i = 0; i < items.size() && d.ok(); ++i) {
+    if (d.num<uint8_t>() != 0) {
+      statuses[i] = d.num<int32_t>();
+      continue;
+    }
+    const uint8_t ncopies = d.num<uint8_t>();
+    placed[i].reserve(ncopies);
+    for (uint8_t c = 0; c < ncopies; ++c) {
+      const uint16_t pi = d.num<uint16_t>();
+      const uint64_t off = d.num<uint64_t>();
+      if (pi >= npools)
+        statuses[i] = static_cast<int32_t>(ErrorCode::PROTOCOL_ERROR);
+      else
+        placed[i].emplace_back(pi, off);
+    }
+  }
+  if (!d.ok()) return Error{ErrorCode::PROTOCOL_ERROR, "bad v2 response"};
+
+  // transfers + digests fan out per item (digest runs while the bytes are
+  // still cache-hot from the memcpy)
+  std::vector<uint64_t> digests(items.size(), 0);
+  {
+    std::atomic<size_t> next{0};
+    const int nthreads = std::max(
+        1, std::min<int>(opts_.io_threads, static_cast<int>(items.size())));
+    std::vector<std::future<void>> futs;
+    for (int t = 0; t < nthreads; ++t)
+      futs.push_back(std::async(std::launch::async, [&] {
+        for (size_t i = next.fetch_add(1); i < items.size();
+             i = next.fetch_add(1)) {
+          if (statuses[i] != 0) continue;
+          for (auto& [pi, off] : placed[i]) {
+            HostPoolRef& pr = pools[pi];
+            if (pr.base) {
+              std::memcpy(pr.base + off, items[i].data, items[i].size);
+            } else {
+              ShardPlacement sp;
+              sp.pool_id = pr.pool_id;
+              sp.offset = off;
+              sp.length = items[i].size;
+              sp.access = pr.access;
+              auto r = write_shard(sp, items[i].data);
+              if (!r.ok()) {
+                statuses[i] = static_cast<int32_t>(r.code());
+                break;
+              }
+            }
+          }
+          if (cfg.checksum && statuses[i] == 0)
+            digests[i] = gpu::checksum_cpu(items[i].data, items[i].size);
+        }
+      }));
+    for (auto& f : futs) f.get();
+  }
+
+  bool all_ok = token != 0;
+  for (auto st : statuses)
+    if (st != 0) { all_ok = false; break; }
+  bool committed = false;
+  if (all_ok) {
+    serde::Enc e2;
+    e2.num<uint64_t>(token);
+    e2.num<uint8_t>(1);  // one-shot: release after commit
+    e2.num<uint32_t>(static_cast<uint32_t>(digests.size()));
+    for (uint64_t dg : digests) e2.num<uint64_t>(dg);
+    committed =
+        meta_call_raw(M::BATCH_COMMIT_TOKEN, e2.buf, opts_.rpc_timeout_ms).ok();
+  }
+  if (!committed) {
+    PutCompleteListMsg completes;
+    std::vector<std::string> cancels;
+    std::vector<size_t> complete_idx;
+    for (size_t i = 0; i < items.size(); ++i) {
+      if (statuses[i] != 0) {
+        if (statuses[i] != static_cast<int32_t>(ErrorCode::OBJECT_EXISTS) &&
+            statuses[i] != static_cast<int32_t>(ErrorCode::NO_SPACE))
+          cancels.push_back(items[i].key);
+        continue;
+      }
+      completes.reqs.push_back(PutCompleteRequest{items[i].key, digests[i]});
+      complete_idx.push_back(i);
+    }
+    if (!completes.reqs.empty()) {
+      auto r = meta_call<PutCompleteListMsg, StatusListMsg>(
+          M::BATCH_PUT_COMPLETE, completes, opts_.rpc_timeout_ms);
+      if (!r.ok()) return r.error();
+      for (size_t j = 0; j < complete_idx.size() && j < r->statuses.size(); ++j)
+        if (r->statuses[j] != 0) statuses[complete_idx[j]] = r->statuses[j];
+    }
+    if (!cancels.empty())
+      meta_call_raw(M::BATCH_PUT_CANCEL, serde::to_bytes(KeysMsg{cancels}),
+                    opts_.rpc_timeout_ms);
+  }
+  return statuses;
+}
+
+Result<std::vector<std::pair<int32_t, std::string>>> Client::batch_get_once_v2(
+    const std::vector<ObjectKey>& keys, bool* fallback) {
+  *fallback = false;
+  serde::Enc req;
+  req.num<uint32_t>(static_cast<uint32_t>(keys.size()));
+  for (auto& k : keys) req.str(k);
+  auto resp =
+      meta_call_raw(M::BATCH_GET_WORKERS2, req.buf, opts_.rpc_timeout_ms);
+  if (!resp.ok()) return resp.error();
+
+  serde::Dec d(resp.value().data(), resp.value().size());
+  const uint16_t npools = d.num<uint16_t>();
+  std::vector<HostPoolRef> pools(npools);
+  for (uint16_t i = 0; i < npools; ++i) {
+    pools[i].pool_id = d.str();
+    pools[i].base = host_pool_base(pools[i].pool_id, &pools[i].access);
+  }
+
+  std::vector<std::pair<int32_t, std::string>> out(keys.size());
+  std::vector<uint64_t> want(keys.size(), 0);
+  std::vector<std::vector<std::pair<uint16_t, uint64_t>>> copies(keys.size());
+  for (size_t i = 0; i < keys.size() && d.ok(); ++i) {
+    if (d.num<uint8_t>() != 0) {
+      out[i].first = d.num<int32_t>();
+      if (out[i].first == static_cast<int32_t>(ErrorCode::NOT_IMPLEMENTED))
+        *fallback = true;  // striped object in the batch: run the v1 path
+      continue;
+    }
+    const uint64_t size = d.num<uint64_t>();
+    want[i] = d.num<uint64_t>();
+    out[i].second.resize(size);
+    const uint8_t ncopies = d.num<uint8_t>();
+    copies[i].reserve(ncopies);
+    for (uint8_t c = 0; c < ncopies; ++c) {
+      const uint16_t pi = d.num<uint16_t>();
+      const uint64_t off = d.num<uint64_t>();
+      if (pi < npools) copies[i].emplace_back(pi, off);
+    }
+  }
+  if (!d.ok()) return Error{ErrorCode::PROTOCOL_ERROR, "bad v2 response"};
+  if (*fallback) return out;
+
+  std::atomic<size_t> next{0};
+  const int nthreads = std::max(
+      1, std::min<int>(opts_.io_threads, static_cast<int>(keys.size())));
+  std::vector<std::future<void>> futs;
+  for (int t = 0; t < nthreads; ++t)
+    futs.push_back(std::async(std::launch::async, [&] {
+      for (size_t i = next.fetch_add(1); i < keys.size();
+           i = next.fetch_add(1)) {
+        if (out[i].first != 0) continue;
+        const uint64_t size = out[i].second.size();
+        Error last{ErrorCode::NO_PLACEMENT, "no copies"};
+        bool done = false;
+        for (auto& [pi, off] : copies[i]) {
+          HostPoolRef& pr = pools[pi];
+          if (pr.base) {
+            std::memcpy(out[i].second.data(), pr.base + off, size);
+            done = true;
+          } else {
+            ShardPlacement sp;
+            sp.pool_id = pr.pool_id;
+            sp.offset = off;
+            sp.length = size;
+            sp.access = pr.access;
+            auto r = read_shard(sp, out[i].second.data());
+            if (r.ok()) done = true;
+            else last = r.error();
+          }
+          if (done && opts_.verify_checksum_on_get && want[i] != 0) {
+            if (gpu::checksum_cpu(out[i].second.data(), size) != want[i]) {
+              done = false;  // corrupt copy: try the next one
+              last = Error{ErrorCode::CHECKSUM_MISMATCH, keys[i]};
+            }
+          }
+          if (done) break;
+        }
+        if (!done) {
+          out[i].first = static_cast<int32_t>(last.code);
+          out[i].second.clear();
+        }
+      }
+    }));
+  for (auto& f : futs) f.get();
+  return out;
+}
+
 Result<std::vector<int32_t>> Client::batch_put_once(const std::vector<PutItem>& items,
                                                const PlacementConfig& cfg) {
+  if (!opts_.force_tcp && cfg.max_workers_per_copy <= 1 && !items.empty())
+    return batch_put_once_v2(items, cfg);
   BatchPutStartRequest breq;
   breq.requests.reserve(items.size());
   for (const auto& it : items)
@@ -739,6 +996,13 @@ Result<std::vector<std::pair<int32_t, std::string>>> Client::batch_get(
 
 Result<std::vector<std::pair<int32_t, std::string>>> Client::batch_get_once(
     const std::vector<ObjectKey>& keys) {
+  if (!opts_.force_tcp && !keys.empty()) {
+    bool fb = false;
+    auto v2 = batch_get_once_v2(keys, &fb);
+    if (v2.ok() && !fb) return v2;
+    if (!v2.ok()) return v2;  // whole-call errors surface (failover retries
+                              // happen one level up in batch_get)
+  }
   auto meta = meta_call<KeysMsg, BatchGetWorkersResponse>(
       M::BATCH_GET_WORKERS, KeysMsg{keys}, opts_.rpc_timeout_ms);
   if (!meta.ok()) return meta.error();

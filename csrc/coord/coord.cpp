@@ -252,6 +252,11 @@ struct KeyReq {
   std::string key;
   BB_FIELDS(key)
 };
+struct PutManyReq {
+  std::vector<KV> puts;
+  std::vector<std::string> dels;
+  BB_FIELDS(puts, dels)
+};
 struct ValueResp {
   std::string value;
   BB_FIELDS(value)
@@ -339,6 +344,16 @@ CoordServer::CoordServer(std::shared_ptr<CoordStore> store) : store_(std::move(s
     if (!r.ok()) return r.error();
     BB_RETURN_IF_ERROR(check_writable(r->second));
     BB_RETURN_IF_ERROR(store_->put(r->first.key, r->first.value, r->first.ttl_ms));
+    return std::string{};
+  });
+  rpc_.register_handler(method::PUT_MANY, [this](const std::string& b, const Ctx&) -> Result<std::string> {
+    auto r = decode_fenced<PutManyReq>(b);
+    if (!r.ok()) return r.error();
+    BB_RETURN_IF_ERROR(check_writable(r->second));
+    for (const auto& kv : r->first.puts)
+      BB_RETURN_IF_ERROR(store_->put(kv.key, kv.value, 0));
+    for (const auto& k : r->first.dels)
+      BB_RETURN_IF_ERROR(store_->del(k));
     return std::string{};
   });
   rpc_.register_handler(method::GET, [this](const std::string& b, const Ctx&) -> Result<std::string> {
@@ -599,6 +614,15 @@ Result<std::string> CoordClient::call_with_retry(uint16_t m, const std::string& 
 
 Result<void> CoordClient::put(const std::string& k, const std::string& v, uint64_t ttl) {
   auto r = call_with_retry(method::PUT, fenced(serde::to_bytes(PutReq{k, v, ttl})));
+  if (!r.ok()) return r.error();
+  return {};
+}
+
+Result<void> CoordClient::put_many(const std::vector<KV>& puts,
+                                   const std::vector<std::string>& dels) {
+  if (puts.empty() && dels.empty()) return {};
+  auto r = call_with_retry(method::PUT_MANY,
+                           fenced(serde::to_bytes(PutManyReq{puts, dels})));
   if (!r.ok()) return r.error();
   return {};
 }

@@ -140,6 +140,16 @@ class Client {
                         const PlacementConfig& cfg);
   Result<std::vector<std::pair<int32_t, std::string>>> batch_get_once(
       const std::vector<ObjectKey>& keys);
+  // compact v2 batch protocol (pool-table responses, fixed-width placements,
+  // token commits) — the host-tier twin of GpuClient's v2 paths. Single-shard
+  // placements only; *fallback signals "run the v1 path instead".
+  Result<std::vector<int32_t>> batch_put_once_v2(
+      const std::vector<PutItem>& items, const PlacementConfig& cfg);
+  Result<std::vector<std::pair<int32_t, std::string>>> batch_get_once_v2(
+      const std::vector<ObjectKey>& keys, bool* fallback);
+  // host-visible base of a pool (same-process host pool or SHM mapping);
+  // nullptr → per-shard write_shard/read_shard fallback
+  uint8_t* host_pool_base(const PoolId& id, AccessInfo* access);
 };
 
 }  // namespace blackbird

@@ -71,6 +71,14 @@ class CoordService {
   virtual Result<void> keep_alive(const std::string& key, uint64_t ttl_ms) = 0;
   virtual Result<uint64_t> watch_prefix(const std::string& prefix, WatchCallback cb) = 0;
   virtual Result<void> unwatch(uint64_t watch_id) = 0;
+  // batched puts (ttl 0) + deletes; remote implementations send ONE round
+  // trip (synchronous durability of a whole commit batch)
+  virtual Result<void> put_many(const std::vector<KV>& puts,
+                                const std::vector<std::string>& dels) {
+    for (const auto& kv : puts) BB_RETURN_IF_ERROR(put(kv.key, kv.value, 0));
+    for (const auto& k : dels) BB_RETURN_IF_ERROR(del(k));
+    return {};
+  }
 };
 
 // ------------------------------------------------------------- the engine
@@ -192,6 +200,7 @@ constexpr uint16_t WATCH = 106;
 constexpr uint16_t UNWATCH = 107;
 constexpr uint16_t DUMP = 108;  // full state incl. TTLs + epoch (replication)
 constexpr uint16_t EPOCH = 109;  // current fencing epoch (u64)
+constexpr uint16_t PUT_MANY = 110;  // batched puts + deletes, one round trip
 }  // namespace method
 
 class CoordServer {
@@ -279,6 +288,8 @@ class CoordClient : public CoordService {
   Result<void> keep_alive(const std::string& k, uint64_t ttl) override;
   Result<uint64_t> watch_prefix(const std::string& p, WatchCallback cb) override;
   Result<void> unwatch(uint64_t id) override;
+  Result<void> put_many(const std::vector<KV>& puts,
+                        const std::vector<std::string>& dels) override;
   // highest leadership epoch observed (sent with every mutation; a stale
   // revived primary rejects + self-fences on seeing a newer one)
   uint64_t observed_epoch() const { return epoch_.load(); }

@@ -143,6 +143,9 @@ class KeystoneService {
   void gc_loop();
   void keepalive_loop();
   void persist_loop();
+  // drain the dirty set synchronously in one coord round trip (durability
+  // of commit/remove acks); call without objects_mu_ held
+  void flush_dirty_now();
   void mark_dirty_locked(const ObjectKey& key, bool removed);
   void setup_watchers();
   void load_existing_state();

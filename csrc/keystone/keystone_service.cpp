@@ -188,8 +188,10 @@ Result<void> KeystoneService::put_complete(
   std::unique_lock lk(objects_mu_);
   auto it = objects_.find(key);
   if (it == objects_.end()) return Error{ErrorCode::OBJECT_NOT_FOUND, key};
-  if (it->second.state == ObjectState::COMMITTED)
+  if (it->second.state == ObjectState::COMMITTED) {
+    if (it->second.checksum == checksum) return {};  // retried commit
     return Error{ErrorCode::INVALID_STATE, "already committed: " + key};
+  }
   it->second.state = ObjectState::COMMITTED;
   it->second.checksum = checksum;
   apply_shard_digests(it->second, checksum, shard_digests);
@@ -197,6 +199,8 @@ Result<void> KeystoneService::put_complete(
   it->second.last_access_ms = it->second.created_ms;
   mark_dirty_locked(key, false);
   bump_view();
+  lk.unlock();
+  flush_dirty_now();
   return {};
 }
 
@@ -407,7 +411,12 @@ std::vector<int32_t> KeystoneService::batch_put_complete(
         continue;
       }
       if (it->second.state == ObjectState::COMMITTED) {
-        out[i] = static_cast<int32_t>(ErrorCode::INVALID_STATE);
+        // idempotent: a commit retried after a leader failover (the first
+        // attempt applied + replicated, the reply was lost) carries the
+        // same digest — report success, not INVALID_STATE
+        out[i] = it->second.checksum == reqs[i].checksum
+                     ? 0
+                     : static_cast<int32_t>(ErrorCode::INVALID_STATE);
         continue;
       }
       it->second.state = ObjectState::COMMITTED;
@@ -420,6 +429,7 @@ std::vector<int32_t> KeystoneService::batch_put_complete(
     }
     if (any) bump_view();
   }
+  flush_dirty_now();
   return out;
 }
 
@@ -528,6 +538,7 @@ Result<void> KeystoneService::commit_token(
     std::lock_guard<std::mutex> g(sessions_mu_);
     put_sessions_.erase(token);
   }
+  flush_dirty_now();
   return {};
 }
 
@@ -604,6 +615,7 @@ std::vector<int32_t> KeystoneService::batch_remove(
   }
   // range frees take only the allocator's own lock — once for the batch
   allocator_.free_batch(to_free);
+  flush_dirty_now();  // deletions durable before the reply (retry-safe)
   return out;
 }
 
@@ -1172,6 +1184,45 @@ void KeystoneService::mark_dirty_locked(const ObjectKey& key, bool removed) {
   if (!config_.persist_objects) return;
   std::lock_guard<std::mutex> g(dirty_mu_);
   dirty_[key] = removed;
+}
+
+// Synchronous durability for commit/remove acknowledgements: drain the dirty
+// set NOW in one coordination round trip (put_many) so an acked mutation
+// survives an immediate leader crash — the 100 ms persist_loop stays as the
+// backstop for maintenance-path mutations (tiering, repair, scrub).
+// Must be called WITHOUT objects_mu_ held (serializes dirty→blob under a
+// shared lock, same order as persist_loop).
+void KeystoneService::flush_dirty_now() {
+  if (!config_.persist_objects || !coord_) return;
+  std::map<ObjectKey, bool> batch;
+  {
+    std::lock_guard<std::mutex> g(dirty_mu_);
+    if (dirty_.empty()) return;
+    batch.swap(dirty_);
+  }
+  const std::string obj_prefix = prefix() + "/objects/";
+  std::vector<coord::KV> puts;
+  std::vector<std::string> dels;
+  {
+    std::shared_lock lk(objects_mu_);
+    for (const auto& [key, removed] : batch) {
+      if (removed) {
+        dels.push_back(obj_prefix + key);
+        continue;
+      }
+      auto it = objects_.find(key);
+      if (it == objects_.end()) continue;
+      puts.push_back(coord::KV{obj_prefix + key, serde::to_bytes(it->second)});
+    }
+  }
+  auto r = coord_->put_many(puts, dels);
+  if (!r.ok()) {
+    // coordination hiccup: put the work back for the async loop to retry
+    BB_LOG(WARN) << "sync persist failed (" << r.error().message
+                 << "); re-queued for async flush";
+    std::lock_guard<std::mutex> g(dirty_mu_);
+    for (auto& [key, removed] : batch) dirty_.emplace(key, removed);
+  }
 }
 
 void KeystoneService::persist_loop() {

@@ -63,9 +63,11 @@ class TestPutLifecycle:
     def test_put_complete_validates_state(self, ks):
         cfg = bb.PlacementConfig()
         ks.put_start("k", 1024, cfg)
-        ks.put_complete("k", 0)
+        ks.put_complete("k", 7)
+        # same-digest duplicate = a commit retried after failover: idempotent
+        ks.put_complete("k", 7)
         with pytest.raises(Exception, match="INVALID_STATE"):
-            ks.put_complete("k", 0)  # double commit rejected (reference didn't)
+            ks.put_complete("k", 8)  # DIFFERENT content double-commit rejected
         with pytest.raises(Exception, match="OBJECT_NOT_FOUND"):
             ks.put_complete("nope", 0)
 
