@@ -225,6 +225,15 @@ PYBIND11_MODULE(_core, m) {
       .def("load", [](coord::CoordStore& s, const std::string& path) {
         unwrap_void(s.load(path));
       })
+      .def("get_prefix",
+           [](coord::CoordStore& s, const std::string& p) {
+             auto r = s.get_prefix(p);
+             if (!r.ok()) throw std::runtime_error(r.error().message);
+             std::vector<std::pair<std::string, py::bytes>> out;
+             for (auto& kv : r.value())
+               out.emplace_back(kv.key, py::bytes(kv.value));
+             return out;
+           })
       .def("dirty", &coord::CoordStore::dirty)
       .def("epoch", &coord::CoordStore::epoch)
       .def("bump_epoch", &coord::CoordStore::bump_epoch);

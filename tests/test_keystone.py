@@ -514,6 +514,58 @@ class TestPutSessions:
             ks.upsert_start_token(tok)
 
 
+class TestDurableAcks:
+    def test_commit_ack_already_in_coordination(self):
+        """With persist_objects, a batch_put/batch_remove ACK means the
+        mutation is ALREADY in coordination (synchronous put_many flush) —
+        no 100 ms async window a leader crash could lose."""
+        store = bb.CoordStore()
+        cs = bb.CoordServer(store)
+        cs.start("127.0.0.1", 0)
+        ep = "127.0.0.1:%d" % cs.port
+        cfg = bb.KeystoneConfig()
+        cfg.listen_address = "127.0.0.1:0"
+        cfg.coord_endpoint = ep
+        cfg.persist_objects = True
+        cfg.gc_interval_ms = 100000
+        srv = bb.create_and_start_keystone(cfg)
+        wc = bb.WorkerConfig()
+        wc.worker_id = "duraw0"
+        wc.coord_endpoint = ep
+        wc.data_listen_address = "127.0.0.1:0"
+        p = bb.PoolConfig()
+        p.pool_id = "durapool"
+        p.storage_class = bb.StorageClass.RAM_CPU
+        p.size_bytes = 64 << 20
+        wc.pools = [p]
+        w = bb.WorkerService(wc)
+        w.initialize()
+        w.start()
+        deadline = time.time() + 5
+        while time.time() < deadline and not srv.service().get_memory_pools():
+            time.sleep(0.02)
+        try:
+            o = bb.ClientOptions()
+            o.keystone_endpoint = srv.endpoint
+            c = bb.Client(o)
+            c.connect()
+            keys = ["dura-%d" % i for i in range(8)]
+            st = c.batch_put([(k, b"x" * 4096) for k in keys])
+            assert all(s == 0 for s in st), st
+            prefix = "/blackbird/clusters/default/objects/"
+            kvs = store.get_prefix(prefix)  # no sleep: durable at ack time
+            assert {k for k, _ in kvs} == {prefix + k for k in keys}
+            st = c.batch_remove(keys)
+            assert all(s == 0 for s in st), st
+            assert store.get_prefix(prefix) == []
+            c.close()
+        finally:
+            w.stop()
+            srv.stop()
+            srv.service().stop()
+            cs.stop()
+
+
 class TestFailoverMidBatch:
     def test_batches_survive_leader_failover_with_zero_errors(self):
         """a keystone leader failover in the MIDDLE of a batch
