@@ -180,6 +180,8 @@ RangeAllocator::allocate_batch(const std::vector<ObjectKey>& keys,
   std::lock_guard<std::mutex> g(mu_);
   const uint32_t replicas = std::max<uint32_t>(cfg.replication, 1);
   const bool striped = cfg.max_workers_per_copy > 1;
+  // pre-size the ledger so a large batch never rehashes mid-insert
+  ledger_.reserve(ledger_.size() + keys.size());
 
   // one candidate scan+sort for the whole batch
   auto cands = candidates_locked(cfg, 1);

@@ -374,6 +374,7 @@ BatchPutStartResponse KeystoneService::batch_put_start(
   }
   {
     std::unique_lock lk(objects_mu_);
+    objects_.reserve(objects_.size() + reqs.size());  // no mid-batch rehash
     for (size_t i = 0; i < reqs.size(); ++i) {
       if (out.items[i].status != 0 || keys[i].empty() || placed[i].first != 0)
         continue;
