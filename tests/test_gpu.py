@@ -2,6 +2,7 @@
 reference (bit-exact), batched scatter/gather copy, HBM backend, and the full
 HBM-tier cluster path (BASELINE config #2)."""
 import os
+import time
 
 import numpy as np
 import pytest
@@ -716,6 +717,76 @@ class TestBatchSessions:
             assert gcl.session_put_steps == 0
             g.free(src)
             c.close()
+        finally:
+            cl.stop()
+
+    def test_concurrent_session_clients(self):
+        """Two GpuClients (own metadata connections, own key spaces) run
+        session-fast-path upsert loops concurrently while a third actor
+        creates/removes keys to bump the placement epoch (invalidating the
+        others' sessions mid-stream). Every batch must succeed and every
+        get must return the bytes that client last put."""
+        import threading
+        cl = Cluster(n_workers=1, pool_bytes=512 * MB,
+                     storage_class=bb.StorageClass.RAM_GPU)
+        g = bb.core.gpu
+        errs = []
+        try:
+            ks = cl.keystone.service()
+
+            def worker(tag, seed):
+                try:
+                    c = cl.client()
+                    gcl = bb.GpuClient(c, 0)
+                    gcl.init()
+                    gcl.set_placement_cache(True)
+                    N, S = 16, 64 * 1024
+                    src = g.malloc(N * S)
+                    dst = g.malloc(N * S)
+                    cfg = bb.PlacementConfig()
+                    cfg.replace = True
+                    pb = bb.make_put_batch(
+                        [("%s%02d" % (tag, i), src + i * S, S)
+                         for i in range(N)])
+                    gb = bb.make_get_batch(
+                        [("%s%02d" % (tag, i), dst + i * S, S)
+                         for i in range(N)])
+                    for step in range(12):
+                        blobs = [os.urandom(S) for _ in range(N)]
+                        for i, b in enumerate(blobs):
+                            g.upload(src + i * S, b)
+                        assert gcl.batch_put_prepared(pb, cfg), (tag, step)
+                        assert gcl.batch_get_prepared(gb), (tag, step)
+                        for i, b in enumerate(blobs):
+                            got = g.download(dst + i * S, S)
+                            assert got == b, (tag, step, i)
+                    # sessions must have engaged despite interference
+                    assert gcl.session_put_steps >= 2, (tag,
+                                                        gcl.session_put_steps)
+                    c.close()
+                    g.free(src)
+                    g.free(dst)
+                except Exception as e:  # surfaced after join
+                    errs.append((tag, repr(e)))
+
+            def intruder():
+                try:
+                    for k in range(30):
+                        ks.put_start("intr%d" % k, 4096, bb.PlacementConfig())
+                        ks.put_complete("intr%d" % k, checksum=k + 1)
+                        ks.remove_object("intr%d" % k)  # bumps epoch
+                        time.sleep(0.01)
+                except Exception as e:
+                    errs.append(("intruder", repr(e)))
+
+            ts = [threading.Thread(target=worker, args=("cA", 1)),
+                  threading.Thread(target=worker, args=("cB", 2)),
+                  threading.Thread(target=intruder)]
+            for t in ts:
+                t.start()
+            for t in ts:
+                t.join()
+            assert not errs, errs
         finally:
             cl.stop()
 
