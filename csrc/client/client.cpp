@@ -463,7 +463,7 @@ Result<void> Client::put_once(const ObjectKey& key, const void* data,
     return xfer.error();
   }
   uint64_t checksum = cfg.checksum ? gpu::checksum_cpu(data, size) : 0;
-  PutCompleteRequest done_req{key, checksum};
+  PutCompleteRequest done_req{key, checksum, {}};
   if (cfg.checksum)
     done_req.shard_digests = host_shard_digests(start->copies, data);
   auto done = meta_call_raw(M::PUT_COMPLETE, serde::to_bytes(done_req),
@@ -722,7 +722,7 @@ Result<std::vector<int32_t>> Client::batch_put_once_v2(
           cancels.push_back(items[i].key);
         continue;
       }
-      completes.reqs.push_back(PutCompleteRequest{items[i].key, digests[i]});
+      completes.reqs.push_back(PutCompleteRequest{items[i].key, digests[i], {}});
       complete_idx.push_back(i);
     }
     if (!completes.reqs.empty()) {
@@ -893,7 +893,7 @@ Result<std::vector<int32_t>> Client::batch_put_once(const std::vector<PutItem>& 
           cancels2.push_back(items[i].key);
           continue;
         }
-        PutCompleteRequest pc{items[i].key, digests[i]};
+        PutCompleteRequest pc{items[i].key, digests[i], {}};
         if (cfg.checksum)
           pc.shard_digests =
               host_shard_digests(start->items[i].copies, items[i].data);
@@ -952,7 +952,7 @@ Result<std::vector<int32_t>> Client::batch_put_once(const std::vector<PutItem>& 
       cancels.push_back(items[i].key);
       continue;
     }
-    PutCompleteRequest pc{items[i].key, digests[i]};
+    PutCompleteRequest pc{items[i].key, digests[i], {}};
     if (cfg.checksum)
       pc.shard_digests =
           host_shard_digests(start->items[i].copies, items[i].data);

@@ -604,11 +604,11 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
     for (size_t j = 0; j < committed_idx.size(); ++j) {
       if (statuses[committed_idx[j]] != 0) continue;  // staged write failed
       completes.reqs.push_back(
-          PutCompleteRequest{items[committed_idx[j]].key, digests[j]});
+          PutCompleteRequest{items[committed_idx[j]].key, digests[j], {}});
     }
     for (size_t j = 0; j < fused_hash_idx.size(); ++j)
       completes.reqs.push_back(
-          PutCompleteRequest{items[fused_hash_idx[j]].key, fused_digests[j]});
+          PutCompleteRequest{items[fused_hash_idx[j]].key, fused_digests[j], {}});
     if (!completes.reqs.empty()) {
       auto r = c_.meta_call<PutCompleteListMsg, StatusListMsg>(
           M::BATCH_PUT_COMPLETE, completes);
@@ -1151,7 +1151,7 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_once(
   PutCompleteListMsg completes;
   std::vector<uint32_t> complete_order;
   for (size_t j = 0; j < committed_idx.size(); ++j) {
-    PutCompleteRequest pc{items[committed_idx[j]].key, digests[j]};
+    PutCompleteRequest pc{items[committed_idx[j]].key, digests[j], {}};
     auto it = shard_digests.find(committed_idx[j]);
     if (it != shard_digests.end()) pc.shard_digests = std::move(it->second);
     completes.reqs.push_back(std::move(pc));
@@ -1159,7 +1159,7 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_once(
   }
   for (size_t j = 0; j < fused_hash_idx.size(); ++j) {
     completes.reqs.push_back(
-        PutCompleteRequest{items[fused_hash_idx[j]].key, fused_digests[j]});
+        PutCompleteRequest{items[fused_hash_idx[j]].key, fused_digests[j], {}});
     complete_order.push_back(fused_hash_idx[j]);
   }
   if (!completes.reqs.empty()) {
