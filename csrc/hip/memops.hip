@@ -443,8 +443,13 @@ Result<void> FusedPutPlan::build(const PutDesc* descs, uint32_t n, int device) {
   if (blocks < 1) blocks = 1;
 
   hipGraph_t graph = nullptr;
+  // Relaxed: every captured op is issued on impl->stream by THIS thread, so
+  // the stricter modes buy nothing — and ROCm's ThreadLocal mode still
+  // fails OTHER threads' legacy-stream hipMemcpy while a capture is open
+  // ("operation would make the legacy stream depend on a capturing blocking
+  // stream"), which a concurrent client must be free to do.
   BB_HIP_TRY(hipStreamBeginCapture(impl->stream,
-                                   hipStreamCaptureModeThreadLocal));
+                                   hipStreamCaptureModeRelaxed));
   hipError_t cap = hipSuccess;
   do {
     if ((cap = hipMemsetAsync(d_out, 0, impl->out_bytes, impl->stream)) !=
