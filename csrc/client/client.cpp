@@ -241,7 +241,7 @@ Result<void> Client::write_shard(const ShardPlacement& s, const void* src) {
         std::memcpy(dst, src, s.length);
         return {};
       }
-      if (hipMemcpy(dst, src, s.length, hipMemcpyHostToDevice) == hipSuccess)
+      if (gpu::copy_sync(dst, src, s.length, hipMemcpyHostToDevice).ok())
         return {};
     } else if (auto* be = LocalPools::inst().backend(s.pool_id)) {
       // unmapped same-process pool (direct-IO NVMe tier): call the backend
@@ -262,10 +262,10 @@ Result<void> Client::write_shard(const ShardPlacement& s, const void* src) {
   if (a.kind == AccessKind::HIP_IPC && !a.ipc_handle_hex.empty() &&
       gpu::available()) {
     if (void* base = mapper_->open_ipc(a.ipc_handle_hex, a.device_id)) {
-      hipError_t e = hipMemcpy(static_cast<uint8_t*>(base) + s.offset, src,
-                               s.length, hipMemcpyHostToDevice);
-      if (e == hipSuccess) return {};
-      BB_LOG(WARN) << "IPC write failed: " << hipGetErrorString(e)
+      auto e = gpu::copy_sync(static_cast<uint8_t*>(base) + s.offset, src,
+                              s.length, hipMemcpyHostToDevice);
+      if (e.ok()) return {};
+      BB_LOG(WARN) << "IPC write failed: " << e.error().message
                    << " — falling back to TCP";
     }
   }
@@ -315,7 +315,7 @@ Result<void> Client::read_shard(const ShardPlacement& s, void* dst) {
         std::memcpy(dst, src2, s.length);
         return {};
       }
-      if (hipMemcpy(dst, src2, s.length, hipMemcpyDeviceToHost) == hipSuccess)
+      if (gpu::copy_sync(dst, src2, s.length, hipMemcpyDeviceToHost).ok())
         return {};
     } else if (auto* be = LocalPools::inst().backend(s.pool_id)) {
       // unmapped same-process pool (direct-IO NVMe tier): read through the
@@ -332,9 +332,9 @@ Result<void> Client::read_shard(const ShardPlacement& s, void* dst) {
   if (a.kind == AccessKind::HIP_IPC && !a.ipc_handle_hex.empty() &&
       gpu::available()) {
     if (void* base = mapper_->open_ipc(a.ipc_handle_hex, a.device_id)) {
-      hipError_t e = hipMemcpy(dst, static_cast<uint8_t*>(base) + s.offset,
-                               s.length, hipMemcpyDeviceToHost);
-      if (e == hipSuccess) return {};
+      auto e = gpu::copy_sync(dst, static_cast<uint8_t*>(base) + s.offset,
+                              s.length, hipMemcpyDeviceToHost);
+      if (e.ok()) return {};
     }
   }
   auto* dc = data_client(a.endpoint);

@@ -171,8 +171,9 @@ Result<void> GpuClient::staged_write(const ShardPlacement& s, const void* dev_sr
   uint64_t done = 0;
   while (done < s.length) {
     uint64_t chunk = std::min(s.length - done, staging_size_);
-    BB_HIP(hipMemcpy(staging_, static_cast<const uint8_t*>(dev_src) + done,
-                     chunk, hipMemcpyDeviceToHost));
+    BB_RETURN_IF_ERROR(gpu::copy_sync(
+        staging_, static_cast<const uint8_t*>(dev_src) + done, chunk,
+        hipMemcpyDeviceToHost));
     ShardPlacement part = s;
     part.offset = s.offset + done;
     part.length = chunk;
@@ -191,8 +192,9 @@ Result<void> GpuClient::staged_read_buf(const ShardPlacement& s, void* dev_dst,
     part.offset = s.offset + done;
     part.length = chunk;
     BB_RETURN_IF_ERROR(c_.read_shard(part, staging));
-    BB_HIP(hipMemcpy(static_cast<uint8_t*>(dev_dst) + done, staging, chunk,
-                     hipMemcpyHostToDevice));
+    BB_RETURN_IF_ERROR(gpu::copy_sync(
+        static_cast<uint8_t*>(dev_dst) + done, staging, chunk,
+        hipMemcpyHostToDevice));
     done += chunk;
   }
   return {};
@@ -204,8 +206,9 @@ Result<void> GpuClient::staged_write_buf(const ShardPlacement& s,
   uint64_t done = 0;
   while (done < s.length) {
     uint64_t chunk = std::min(s.length - done, staging_size);
-    BB_HIP(hipMemcpy(staging, static_cast<const uint8_t*>(dev_src) + done,
-                     chunk, hipMemcpyDeviceToHost));
+    BB_RETURN_IF_ERROR(gpu::copy_sync(
+        staging, static_cast<const uint8_t*>(dev_src) + done, chunk,
+        hipMemcpyDeviceToHost));
     ShardPlacement part = s;
     part.offset = s.offset + done;
     part.length = chunk;

@@ -246,12 +246,12 @@ Result<void> GpuCopier::copy_batch(const gpu::CopyDesc* descs, uint32_t n) {
 }
 
 Result<void> GpuCopier::to_transport(void* dst, const void* src, uint64_t n) {
-  BB_HIP(hipMemcpy(dst, src, n, hipMemcpyHostToDevice));
+  BB_RETURN_IF_ERROR(gpu::copy_sync(dst, src, n, hipMemcpyHostToDevice));
   return {};
 }
 
 Result<void> GpuCopier::from_transport(void* dst, const void* src, uint64_t n) {
-  BB_HIP(hipMemcpy(dst, src, n, hipMemcpyDeviceToHost));
+  BB_RETURN_IF_ERROR(gpu::copy_sync(dst, src, n, hipMemcpyDeviceToHost));
   return {};
 }
 

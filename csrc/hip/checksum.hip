@@ -281,11 +281,11 @@ Result<void> mfma_i8_probe(const int8_t* host_a, const int8_t* host_b,
   BB_HIP_TRY(hipMalloc(&dA, 1024));
   BB_HIP_TRY(hipMalloc(&dB, 1024));
   BB_HIP_TRY(hipMalloc(&dC, 1024 * sizeof(int32_t)));
-  BB_HIP_TRY(hipMemcpy(dA, host_a, 1024, hipMemcpyHostToDevice));
-  BB_HIP_TRY(hipMemcpy(dB, host_b, 1024, hipMemcpyHostToDevice));
+  BB_RETURN_IF_ERROR(copy_sync(dA, host_a, 1024, hipMemcpyHostToDevice));
+  BB_RETURN_IF_ERROR(copy_sync(dB, host_b, 1024, hipMemcpyHostToDevice));
   mfma_i8_probe_kernel<<<1, 64>>>(dA, dB, dC);
   BB_HIP_TRY(hipGetLastError());
-  BB_HIP_TRY(hipMemcpy(host_c, dC, 1024 * sizeof(int32_t), hipMemcpyDeviceToHost));
+  BB_RETURN_IF_ERROR(copy_sync(host_c, dC, 1024 * sizeof(int32_t), hipMemcpyDeviceToHost));
   BB_HIP_TRY(hipFree(dA));
   BB_HIP_TRY(hipFree(dB));
   BB_HIP_TRY(hipFree(dC));

@@ -197,6 +197,28 @@ Result<uint64_t> verify_pattern(const void* dev_ptr, uint64_t nbytes, uint64_t s
 
 namespace blackbird::gpu {
 
+namespace {
+// per-thread non-blocking transfer stream (legacy-stream-free library: a
+// null-stream op in ANY thread fails while a graph capture is open)
+struct TlXferStream {
+  hipStream_t s = nullptr;
+  ~TlXferStream() {
+    if (s) (void)hipStreamDestroy(s);
+  }
+};
+thread_local TlXferStream g_xfer;
+}  // namespace
+
+Result<void> copy_sync(void* dst, const void* src, uint64_t nbytes, int kind) {
+  if (nbytes == 0) return {};
+  if (!g_xfer.s)
+    BB_HIP_TRY(hipStreamCreateWithFlags(&g_xfer.s, hipStreamNonBlocking));
+  BB_HIP_TRY(hipMemcpyAsync(dst, src, nbytes,
+                            static_cast<hipMemcpyKind>(kind), g_xfer.s));
+  BB_HIP_TRY(hipStreamSynchronize(g_xfer.s));
+  return {};
+}
+
 Result<uint64_t> device_malloc(uint64_t nbytes, int device) {
   BB_HIP_TRY(hipSetDevice(device));
   void* p = nullptr;
@@ -210,15 +232,13 @@ Result<void> device_free(uint64_t ptr) {
 }
 
 Result<void> upload(uint64_t dst_dev, const void* src, uint64_t nbytes) {
-  BB_HIP_TRY(hipMemcpy(reinterpret_cast<void*>(dst_dev), src, nbytes,
-                       hipMemcpyHostToDevice));
-  return {};
+  return copy_sync(reinterpret_cast<void*>(dst_dev), src, nbytes,
+                   hipMemcpyHostToDevice);
 }
 
 Result<void> download(void* dst, uint64_t src_dev, uint64_t nbytes) {
-  BB_HIP_TRY(hipMemcpy(dst, reinterpret_cast<const void*>(src_dev), nbytes,
-                       hipMemcpyDeviceToHost));
-  return {};
+  return copy_sync(dst, reinterpret_cast<const void*>(src_dev), nbytes,
+                   hipMemcpyDeviceToHost);
 }
 
 Result<void> sync() {
@@ -433,10 +453,10 @@ Result<void> FusedPutPlan::build(const PutDesc* descs, uint32_t n, int device) {
       static_cast<uint8_t*>(impl->dev) + segs_bytes);
   auto* d_out = reinterpret_cast<unsigned long long*>(
       static_cast<uint8_t*>(impl->dev) + segs_bytes + prefix_bytes);
-  BB_HIP_TRY(hipMemcpy(d_segs, h_segs.data(), segs_bytes,
-                       hipMemcpyHostToDevice));
-  BB_HIP_TRY(hipMemcpy(d_prefix, h_prefix.data(), prefix_bytes,
-                       hipMemcpyHostToDevice));
+  BB_RETURN_IF_ERROR(copy_sync(d_segs, h_segs.data(), segs_bytes,
+                               hipMemcpyHostToDevice));
+  BB_RETURN_IF_ERROR(copy_sync(d_prefix, h_prefix.data(), prefix_bytes,
+                               hipMemcpyHostToDevice));
 
   uint64_t waves = (total + 1) / 2;
   uint64_t blocks = std::min<uint64_t>((waves + 3) / 4, 4096);

@@ -111,6 +111,13 @@ Result<void> fill_pattern(void* dev_ptr, uint64_t nbytes, uint64_t seed,
 Result<uint64_t> verify_pattern(const void* dev_ptr, uint64_t nbytes, uint64_t seed,
                                 hipStream_t stream);  // returns #mismatched u64
 
+// Synchronous copy on a per-thread NON-BLOCKING stream. The library never
+// issues work on the legacy (null) stream: ROCm fails legacy-stream ops in
+// EVERY thread while any hipGraph capture is open (sessions capture their
+// step graphs lazily), so all blocking copies route through here. kind is a
+// hipMemcpyKind (int to keep hip types out of this header).
+Result<void> copy_sync(void* dst, const void* src, uint64_t nbytes, int kind);
+
 // Raw device memory helpers (Python test harness / bench plumbing).
 Result<uint64_t> device_malloc(uint64_t nbytes, int device);
 Result<void> device_free(uint64_t ptr);
