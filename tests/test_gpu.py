@@ -287,6 +287,37 @@ class TestRcclEngine:
 
 
 class TestGpuClientPaths:
+    def test_jumbo_object_roundtrip(self):
+        """One 200 MiB object through the fused copy+digest path: bytes land
+        exactly, the kernel digest matches the bit-exact CPU reference, and
+        the verified get passes. (288 GB HBM pools make multi-hundred-MiB
+        objects a normal case, not an edge.)"""
+        S = 200 * MB
+        cl = Cluster(n_workers=1, pool_bytes=512 * MB,
+                     storage_class=bb.StorageClass.RAM_GPU)
+        g = bb.core.gpu
+        try:
+            c = cl.client()
+            gcl = bb.GpuClient(c, 0)
+            gcl.init()
+            src = g.malloc(S)
+            dst = g.malloc(S)
+            g.fill_pattern(src, S, seed=4242)
+            g.sync()
+            assert gcl.batch_put_device([("jumbo", src, S)]) == [0]
+            assert gcl.batch_get_device([("jumbo", dst, S)], verify=True) == [0]
+            g.sync()
+            assert g.verify_pattern(dst, S, seed=4242) == 0
+            blob = g.download(src, S)
+            info = cl.keystone.service().get_workers("jumbo")
+            assert info.size == S
+            assert info.checksum == g.checksum_cpu(blob)
+            c.close()
+            g.free(src)
+            g.free(dst)
+        finally:
+            cl.stop()
+
     def test_device_put_get_fused_and_verified(self):
         """GpuClient batch_put_device (fused copy+digest kernel path) and
         get_device: bytes land correctly, digests match the CPU reference,
