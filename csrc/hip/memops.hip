@@ -8,6 +8,7 @@
 // so a batch of any shape fills 256 CUs; descriptors are binary-searched per
 // chunk (L2-resident, lane-uniform per wave).
 #include <hip/hip_runtime.h>
+#include <cstdlib>
 #include <cstring>
 
 #include <memory>
@@ -211,6 +212,11 @@ thread_local TlXferStream g_xfer;
 
 Result<void> copy_sync(void* dst, const void* src, uint64_t nbytes, int kind) {
   if (nbytes == 0) return {};
+  static const bool legacy = std::getenv("BB_COPY_LEGACY") != nullptr;
+  if (legacy) {  // A/B diagnostics only: null-stream copies break captures
+    BB_HIP_TRY(hipMemcpy(dst, src, nbytes, static_cast<hipMemcpyKind>(kind)));
+    return {};
+  }
   if (!g_xfer.s)
     BB_HIP_TRY(hipStreamCreateWithFlags(&g_xfer.s, hipStreamNonBlocking));
   BB_HIP_TRY(hipMemcpyAsync(dst, src, nbytes,
