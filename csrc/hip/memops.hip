@@ -11,7 +11,9 @@
 #include <cstdlib>
 #include <cstring>
 
+#include <atomic>
 #include <memory>
+#include <mutex>
 #include <vector>
 
 #include "blackbird/gpu/digest_spec.h"
@@ -199,15 +201,32 @@ Result<uint64_t> verify_pattern(const void* dev_ptr, uint64_t nbytes, uint64_t s
 namespace blackbird::gpu {
 
 namespace {
-// per-thread non-blocking transfer stream (legacy-stream-free library: a
-// null-stream op in ANY thread fails while a graph capture is open)
-struct TlXferStream {
-  hipStream_t s = nullptr;
-  ~TlXferStream() {
-    if (s) (void)hipStreamDestroy(s);
+// Shared pool of pre-created non-blocking transfer streams. The library
+// never touches the legacy (null) stream — a null-stream op in ANY thread
+// fails while a hipGraph capture is open — but streams must NOT be
+// per-thread: the staged fan-out paths run on short-lived std::async
+// threads, and creating/destroying a stream per thread costs milliseconds
+// on ROCm (measured 9× on the NVMe get leg). Round-robin over a fixed pool
+// instead; a per-stream mutex keeps each async+sync pair atomic.
+constexpr int kXferStreams = 8;
+struct XferPool {
+  hipStream_t s[kXferStreams] = {};
+  std::mutex mu[kXferStreams];
+  std::atomic<uint32_t> next{0};
+  std::once_flag once;
+  hipError_t init_rc = hipSuccess;
+
+  hipError_t ensure() {
+    std::call_once(once, [this] {
+      for (int i = 0; i < kXferStreams; ++i) {
+        init_rc = hipStreamCreateWithFlags(&s[i], hipStreamNonBlocking);
+        if (init_rc != hipSuccess) return;
+      }
+    });
+    return init_rc;
   }
 };
-thread_local TlXferStream g_xfer;
+XferPool g_xfer;
 }  // namespace
 
 Result<void> copy_sync(void* dst, const void* src, uint64_t nbytes, int kind) {
@@ -217,11 +236,12 @@ Result<void> copy_sync(void* dst, const void* src, uint64_t nbytes, int kind) {
     BB_HIP_TRY(hipMemcpy(dst, src, nbytes, static_cast<hipMemcpyKind>(kind)));
     return {};
   }
-  if (!g_xfer.s)
-    BB_HIP_TRY(hipStreamCreateWithFlags(&g_xfer.s, hipStreamNonBlocking));
+  BB_HIP_TRY(g_xfer.ensure());
+  const uint32_t i = g_xfer.next.fetch_add(1) % kXferStreams;
+  std::lock_guard<std::mutex> g(g_xfer.mu[i]);
   BB_HIP_TRY(hipMemcpyAsync(dst, src, nbytes,
-                            static_cast<hipMemcpyKind>(kind), g_xfer.s));
-  BB_HIP_TRY(hipStreamSynchronize(g_xfer.s));
+                            static_cast<hipMemcpyKind>(kind), g_xfer.s[i]));
+  BB_HIP_TRY(hipStreamSynchronize(g_xfer.s[i]));
   return {};
 }
 
