@@ -62,6 +62,8 @@ GpuClient::~GpuClient() {
     (void)hipSetDevice(device_);
     for (auto& s : streams_)
       if (s) (void)hipStreamDestroy(s);
+    for (auto& s : fan_streams_)
+      if (s) (void)hipStreamDestroy(s);
     if (staging_) (void)hipHostFree(staging_);
     for (void* p : staging_pool_) (void)hipHostFree(p);
   }
@@ -93,6 +95,8 @@ Result<void> GpuClient::init() {
   if (!gpu::available()) return Error{ErrorCode::NO_GPU, "no MI355X visible"};
   BB_HIP(hipSetDevice(device_));
   for (auto& s : streams_)
+    BB_HIP(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+  for (auto& s : fan_streams_)
     BB_HIP(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
   BB_HIP(hipHostMalloc(&staging_, staging_size_, hipHostMallocDefault));
   initialized_ = true;
@@ -222,23 +226,74 @@ Result<void> GpuClient::staged_write_many(
     const std::vector<std::pair<ShardPlacement, const void*>>& work) {
   if (work.empty()) return {};
   if (work.size() == 1) return staged_write(work[0].first, work[0].second);
-  const int nthreads = std::min<int>(kFanThreads,
-                                     static_cast<int>(work.size()));
+  const uint64_t half = kFanBuf / 2;
+  // items that fit a half-buffer ride a 1-deep D2H pipeline (the next
+  // item's device→pinned copy flies while the current one is written to
+  // the backend); oversized items take the chunked path
+  std::vector<size_t> small, large;
+  for (size_t i = 0; i < work.size(); ++i)
+    (work[i].first.length <= half ? small : large).push_back(i);
+  const int nthreads =
+      std::min<int>(kFanThreads, static_cast<int>(work.size()));
   std::atomic<size_t> next{0};
+  std::atomic<size_t> lnext{0};
   std::vector<std::future<Result<void>>> futs;
   for (int t = 0; t < nthreads; ++t)
-    futs.push_back(std::async(std::launch::async, [&]() -> Result<void> {
+    futs.push_back(std::async(std::launch::async, [&, t]() -> Result<void> {
       void* buf = acquire_staging_buf();
       if (!buf) return Error{ErrorCode::HIP_ERROR, "staging alloc"};
+      uint8_t* halves[2] = {static_cast<uint8_t*>(buf),
+                            static_cast<uint8_t*>(buf) + half};
+      hipStream_t st = fan_streams_[t % kFanThreads];
+      hipEvent_t ev[2] = {};
       Result<void> rc{};
-      for (size_t i = next.fetch_add(1); i < work.size();
-           i = next.fetch_add(1)) {
-        auto r = staged_write_buf(work[i].first, work[i].second, buf, kFanBuf);
-        if (!r.ok()) {
-          rc = r;
-          break;
+      auto fail = [&](Result<void> r) { if (rc.ok()) rc = r; };
+      for (auto& e : ev)
+        if (hipEventCreateWithFlags(&e, hipEventDisableTiming) != hipSuccess)
+          fail(Error{ErrorCode::HIP_ERROR, "event create"});
+      if (rc.ok()) {
+        size_t inflight = SIZE_MAX;  // index whose D2H is on halves[ib^1]
+        int ib = 0;
+        for (;;) {
+          size_t k = next.fetch_add(1);
+          const bool have_new = k < small.size();
+          if (have_new) {
+            auto& [sp, src] = work[small[k]];
+            if (hipMemcpyAsync(halves[ib], src, sp.length,
+                               hipMemcpyDeviceToHost, st) != hipSuccess ||
+                hipEventRecord(ev[ib], st) != hipSuccess) {
+              fail(Error{ErrorCode::HIP_ERROR, "staged D2H"});
+              break;
+            }
+          }
+          if (inflight != SIZE_MAX) {
+            // wait only for the PREVIOUS D2H (FIFO stream: its event fired
+            // before the new copy completes), then write it out while the
+            // new copy flies
+            if (hipEventSynchronize(ev[ib ^ 1]) != hipSuccess) {
+              fail(Error{ErrorCode::HIP_ERROR, "staged D2H sync"});
+              break;
+            }
+            auto r = c_.write_shard(work[inflight].first, halves[ib ^ 1]);
+            if (!r.ok()) {
+              fail(r);
+              break;
+            }
+          }
+          if (!have_new) break;
+          inflight = small[k];
+          ib ^= 1;
+        }
+        // chunked path for oversized items
+        for (size_t k = lnext.fetch_add(1); rc.ok() && k < large.size();
+             k = lnext.fetch_add(1)) {
+          auto r = staged_write_buf(work[large[k]].first, work[large[k]].second,
+                                    buf, kFanBuf);
+          if (!r.ok()) fail(r);
         }
       }
+      for (auto& e : ev)
+        if (e) (void)hipEventDestroy(e);
       release_staging_buf(buf);
       return rc;
     }));

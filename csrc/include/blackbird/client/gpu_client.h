@@ -196,6 +196,9 @@ class GpuClient {
       async_;
   static constexpr int kStreams = 7;  // one per xGMI link
   hipStream_t streams_[kStreams] = {};
+  // dedicated streams for the staged fan-out threads (D2H pipelining that
+  // never touches the batch streams or the legacy stream)
+  hipStream_t fan_streams_[8] = {};
   void* staging_ = nullptr;  // pinned bounce buffer for TCP/SHM pools
   uint64_t staging_size_ = 64ull << 20;
   std::mutex staging_mu_;  // async batches share the bounce buffer
