@@ -522,7 +522,11 @@ Result<void> KeystoneService::commit_token(
     if (s->metas[i]->size != s->sizes[i])
       return Error{ErrorCode::SESSION_STALE, "object shape changed"};
   const bool persist = config_.persist_objects;
-  for (size_t i = 0; i < s->metas.size(); ++i) {
+  const size_t n = s->metas.size();
+  for (size_t i = 0; i < n; ++i) {
+    // metas are scattered map nodes: prefetch ahead — this loop is in the
+    // hot warm-step path (one commit per session put step)
+    if (i + 8 < n) __builtin_prefetch(s->metas[i + 8], 1, 1);
     ObjectMeta* m = s->metas[i];
     m->state = ObjectState::COMMITTED;
     m->checksum = digests[i];
