@@ -482,7 +482,14 @@ Result<void> KeystoneService::upsert_start_token(uint64_t token) {
       return Error{ErrorCode::SESSION_STALE, "unknown put session"};
     s = it->second;
   }
-  std::unique_lock lk(objects_mu_);
+  // SHARED lock: session steps from N ranks run concurrently (the warm
+  // plane would otherwise convoy on one mutex at 8 ranks × 2 lanes).
+  // Removal/GC/eviction hold the lock exclusively, so the session's meta
+  // pointers cannot be invalidated inside this section; mutated fields go
+  // through atomic_ref because another session on the SAME keys may write
+  // them concurrently (same-key concurrent upserts are last-writer-wins,
+  // consistent with in-place data overwrites being racy by design).
+  std::shared_lock lk(objects_mu_);
   if (s->epoch != placement_epoch_)
     return Error{ErrorCode::SESSION_STALE, "placements changed"};
   // validate ALL before flipping ANY (all-or-nothing)
@@ -496,7 +503,9 @@ Result<void> KeystoneService::upsert_start_token(uint64_t token) {
   // PENDING pins the placements: tiering/eviction/repair/scrub only touch
   // COMMITTED objects, so the client's one-sided writes land in ranges that
   // cannot move underneath them
-  for (auto* m : s->metas) m->state = ObjectState::PENDING;
+  for (auto* m : s->metas)
+    std::atomic_ref<ObjectState>(m->state)
+        .store(ObjectState::PENDING, std::memory_order_relaxed);
   return {};
 }
 
@@ -515,7 +524,8 @@ Result<void> KeystoneService::commit_token(
   if (digests.size() != s->metas.size())
     return Error{ErrorCode::INVALID_ARGUMENT, "digest count mismatch"};
   const uint64_t now = now_ms();
-  std::unique_lock lk(objects_mu_);
+  // SHARED lock — see upsert_start_token for the concurrency argument
+  std::shared_lock lk(objects_mu_);
   if (s->epoch != placement_epoch_)
     return Error{ErrorCode::SESSION_STALE, "placements changed"};
   for (size_t i = 0; i < s->metas.size(); ++i)
@@ -528,12 +538,17 @@ Result<void> KeystoneService::commit_token(
     // hot warm-step path (one commit per session put step)
     if (i + 8 < n) __builtin_prefetch(s->metas[i + 8], 1, 1);
     ObjectMeta* m = s->metas[i];
-    m->state = ObjectState::COMMITTED;
-    m->checksum = digests[i];
+    std::atomic_ref<uint64_t>(m->checksum)
+        .store(digests[i], std::memory_order_relaxed);
     // sessions cover single-copy single-shard objects by construction
-    m->copies[0].shards[0].digest = digests[i];
-    m->created_ms = now;  // TTL restarts at commit, as in put_complete
-    m->last_access_ms = now;
+    std::atomic_ref<uint64_t>(m->copies[0].shards[0].digest)
+        .store(digests[i], std::memory_order_relaxed);
+    std::atomic_ref<uint64_t>(m->created_ms)
+        .store(now, std::memory_order_relaxed);  // TTL restarts at commit
+    std::atomic_ref<uint64_t>(m->last_access_ms)
+        .store(now, std::memory_order_relaxed);
+    std::atomic_ref<ObjectState>(m->state)
+        .store(ObjectState::COMMITTED, std::memory_order_release);
     if (persist) mark_dirty_locked(m->key, false);
   }
   bump_view();
