@@ -573,34 +573,75 @@ std::vector<int32_t> KeystoneService::batch_put_cancel(
 
 BatchGetWorkersResponse KeystoneService::batch_get_workers(
     const std::vector<ObjectKey>& keys) {
-  // single lock pass (the per-key path pays a unique_lock per object)
+  // SHARED lock: get batches from N ranks run concurrently with each other
+  // and with session commits. Fields a concurrent commit mutates (state,
+  // checksum, shard digest, access times) are read through atomic_ref; the
+  // placement STRUCTURE (strings, offsets, vector shapes) only changes
+  // under the exclusive lock, so plain reads of those are safe. Expired
+  // objects are collected and removed in a rare second exclusive pass.
   BatchGetWorkersResponse out;
   out.items.resize(keys.size());
   const uint64_t now = now_ms();
-  std::unique_lock lk(objects_mu_);
-  for (size_t i = 0; i < keys.size(); ++i) {
-    auto& item = out.items[i];
-    auto it = objects_.find(keys[i]);
-    if (it == objects_.end()) {
-      item.status = static_cast<int32_t>(ErrorCode::OBJECT_NOT_FOUND);
-      continue;
+  std::vector<size_t> expired_idx;
+  {
+    std::shared_lock lk(objects_mu_);
+    for (size_t i = 0; i < keys.size(); ++i) {
+      auto& item = out.items[i];
+      auto it = objects_.find(keys[i]);
+      if (it == objects_.end()) {
+        item.status = static_cast<int32_t>(ErrorCode::OBJECT_NOT_FOUND);
+        continue;
+      }
+      auto& meta = it->second;
+      const auto state = std::atomic_ref<ObjectState>(meta.state)
+                             .load(std::memory_order_acquire);
+      if (state != ObjectState::COMMITTED) {
+        item.status = static_cast<int32_t>(ErrorCode::OBJECT_NOT_COMMITTED);
+        continue;
+      }
+      const uint64_t created = std::atomic_ref<uint64_t>(meta.created_ms)
+                                   .load(std::memory_order_relaxed);
+      if (meta.ttl_ms > 0 && now > created + meta.ttl_ms) {
+        expired_idx.push_back(i);
+        item.status = static_cast<int32_t>(ErrorCode::OBJECT_EXPIRED);
+        continue;
+      }
+      std::atomic_ref<uint64_t>(meta.last_access_ms)
+          .store(now, std::memory_order_relaxed);
+      std::atomic_ref<uint32_t>(meta.access_count)
+          .fetch_add(1, std::memory_order_relaxed);
+      item.status = 0;
+      item.info.size = meta.size;
+      item.info.checksum = std::atomic_ref<uint64_t>(meta.checksum)
+                               .load(std::memory_order_relaxed);
+      item.info.copies.reserve(meta.copies.size());
+      for (auto& c : meta.copies) {
+        CopyPlacement cp;
+        cp.copy_index = c.copy_index;
+        cp.shards.reserve(c.shards.size());
+        for (auto& sh : c.shards) {
+          ShardPlacement sp;
+          sp.pool_id = sh.pool_id;
+          sp.worker_id = sh.worker_id;
+          sp.storage_class = sh.storage_class;
+          sp.offset = sh.offset;
+          sp.length = sh.length;
+          sp.digest = std::atomic_ref<uint64_t>(sh.digest)
+                          .load(std::memory_order_relaxed);
+          sp.access = sh.access;
+          cp.shards.push_back(std::move(sp));
+        }
+        item.info.copies.push_back(std::move(cp));
+      }
     }
-    auto& meta = it->second;
-    if (meta.state != ObjectState::COMMITTED) {
-      item.status = static_cast<int32_t>(ErrorCode::OBJECT_NOT_COMMITTED);
-      continue;
+  }
+  if (!expired_idx.empty()) {
+    std::unique_lock lk(objects_mu_);
+    for (size_t i : expired_idx) {
+      auto it = objects_.find(keys[i]);
+      if (it != objects_.end() && it->second.expired(now))
+        remove_object_locked(keys[i]);
     }
-    if (meta.expired(now)) {
-      remove_object_locked(keys[i]);
-      item.status = static_cast<int32_t>(ErrorCode::OBJECT_EXPIRED);
-      continue;
-    }
-    meta.last_access_ms = now;
-    meta.access_count++;
-    item.status = 0;
-    item.info.copies = meta.copies;
-    item.info.size = meta.size;
-    item.info.checksum = meta.checksum;
   }
   return out;
 }
