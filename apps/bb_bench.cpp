@@ -31,6 +31,7 @@ int main(int argc, char** argv) {
   int iters = 32;
   int batch = 0;   // 0 = single-object mode
   int gpu = -1;    // ≥0 = GPU mode: device buffers + GpuClient fused path
+  bool upsert = true;  // steady-state mode: in-place upserts + batch sessions
   PlacementConfig pcfg;
   for (int i = 1; i < argc; ++i) {
     std::string a = argv[i];
@@ -42,10 +43,13 @@ int main(int argc, char** argv) {
     else if (a == "--replication") pcfg.replication = atoi(next().c_str());
     else if (a == "--stripe") pcfg.max_workers_per_copy = atoi(next().c_str());
     else if (a == "--gpu") gpu = atoi(next().c_str());
+    else if (a == "--no-upsert") upsert = false;
     else if (a == "--help" || a == "-h") {
       std::cout << "bb_bench [--keystone H:P] [--size BYTES] [--iters N]\n"
                    "         [--batch N] [--replication N] [--stripe N]\n"
-                   "         [--gpu DEV]   device buffers + fused GPU data plane\n";
+                   "         [--gpu DEV]   device buffers + fused GPU data plane\n"
+                   "         [--no-upsert] fresh keys + removes each iter (no\n"
+                   "                       session fast path)\n";
       return 0;
     }
   }
@@ -74,6 +78,9 @@ int main(int argc, char** argv) {
       return 1;
     }
     gc.set_placement_cache(true);  // digest-verified RPC-free gets
+    if (upsert && pcfg.replication <= 1) pcfg.replace = true;
+    GpuClient::BatchPutSession psess;
+    GpuClient::BatchGetSession gsess;
     auto src = gpu::device_malloc(size * n, gpu);
     auto dst = gpu::device_malloc(size * n, gpu);
     if (!src.ok() || !dst.ok()) {
@@ -94,7 +101,7 @@ int main(int argc, char** argv) {
     }
     for (int it = 0; it < iters; ++it) {
       auto t0 = Clock::now();
-      auto pr = gc.batch_put_device(puts, pcfg);
+      auto pr = gc.batch_put_device(puts, pcfg, pcfg.replace ? &psess : nullptr);
       if (!pr.ok()) {
         std::cerr << "gpu batch_put failed: " << pr.message() << "\n";
         return 1;
@@ -107,7 +114,7 @@ int main(int argc, char** argv) {
       put_ms.push_back(ms_since(t0));
       put_bytes += static_cast<double>(size) * n;
       t0 = Clock::now();
-      auto gr = gc.batch_get_device(gets);
+      auto gr = gc.batch_get_device(gets, false, pcfg.replace ? &gsess : nullptr);
       if (!gr.ok()) {
         std::cerr << "gpu batch_get failed: " << gr.message() << "\n";
         return 1;
@@ -119,8 +126,9 @@ int main(int argc, char** argv) {
         }
       get_ms.push_back(ms_since(t0));
       get_bytes += static_cast<double>(size) * n;
-      client.batch_remove(keys);
+      if (!pcfg.replace) client.batch_remove(keys);
     }
+    if (pcfg.replace) client.batch_remove(keys);
     // spot-verify last round's payloads against the known fill pattern
     for (int b = 0; b < n && b < 4; ++b) {
       auto bad = gpu::verify_pattern(
