@@ -420,8 +420,14 @@ Result<void> GpuClient::session_kernel(Sess* sess, uint64_t* digests) {
     auto plan = std::make_shared<gpu::FusedPutPlan>();
     auto rb = plan->build(sess->descs.data(),
                           static_cast<uint32_t>(sess->descs.size()), device_);
-    if (rb.ok()) sess->plan = std::move(plan);
-    else sess->plan_failed = true;  // capture unsupported: launch path
+    if (rb.ok()) {
+      sess->plan = std::move(plan);
+    } else {
+      sess->plan_failed = true;  // capture unsupported: launch path
+      (void)hipGetLastError();   // clear any sticky launch error
+      BB_LOG(WARN) << "session graph capture unavailable ("
+                   << rb.error().message << "); using per-op launches";
+    }
   }
   if (sess->plan) {
     auto r = sess->plan->run(digests);
