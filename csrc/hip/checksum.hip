@@ -209,7 +209,8 @@ Result<void> checksum_async(const void* dev_ptr, uint64_t nbytes, uint64_t* dev_
     bbhash64_kernel<<<blocks, kBlock, 0, stream>>>(
         static_cast<const uint8_t*>(dev_ptr), nbytes,
         reinterpret_cast<unsigned long long*>(dev_out));
-    BB_HIP_TRY(hipGetLastError());
+    if (hipError_t _le = hipGetLastError(); _le != hipSuccess)
+      return hip_error(_le, "bbhash64_kernel launch");
   }
   return {};
 }
@@ -263,10 +264,12 @@ Result<void> checksum_batch(const void* const* dev_ptrs, const uint64_t* sizes,
     const int blocks = pick_grid(total);
     bbhash64_batch_kernel<<<blocks, kBlock, 0, stream>>>(d_objs, d_prefix, n,
                                                          total, d_out);
-    BB_HIP_TRY(hipGetLastError());
+    if (hipError_t _le = hipGetLastError(); _le != hipSuccess)
+      return hip_error(_le, "bbhash64_batch_kernel launch");
   }
   bbhash64_finalize_kernel<<<(n + 255) / 256, 256, 0, stream>>>(d_objs, n, d_out);
-  BB_HIP_TRY(hipGetLastError());
+  if (hipError_t _le = hipGetLastError(); _le != hipSuccess)
+    return hip_error(_le, "bbhash64_finalize_kernel launch");
   BB_HIP_TRY(hipMemcpyAsync(h_out, d_out, out_bytes, hipMemcpyDeviceToHost, stream));
   BB_HIP_TRY(hipStreamSynchronize(stream));
   std::memcpy(out_digests, h_out, out_bytes);
@@ -284,7 +287,8 @@ Result<void> mfma_i8_probe(const int8_t* host_a, const int8_t* host_b,
   BB_RETURN_IF_ERROR(copy_sync(dA, host_a, 1024, hipMemcpyHostToDevice));
   BB_RETURN_IF_ERROR(copy_sync(dB, host_b, 1024, hipMemcpyHostToDevice));
   mfma_i8_probe_kernel<<<1, 64>>>(dA, dB, dC);
-  BB_HIP_TRY(hipGetLastError());
+  if (hipError_t _le = hipGetLastError(); _le != hipSuccess)
+    return hip_error(_le, "mfma_i8_probe_kernel launch");
   BB_RETURN_IF_ERROR(copy_sync(host_c, dC, 1024 * sizeof(int32_t), hipMemcpyDeviceToHost));
   BB_HIP_TRY(hipFree(dA));
   BB_HIP_TRY(hipFree(dB));

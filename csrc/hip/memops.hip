@@ -162,7 +162,8 @@ Result<void> batched_copy(const CopyDesc* descs, uint32_t n, hipStream_t stream)
   // one wave per chunk → want total waves ≈ total chunks
   const int blocks = grid_for(total * 64);
   batched_copy_kernel<<<blocks, kBlock, 0, stream>>>(d_segs, d_prefix, n, total);
-  BB_HIP_TRY(hipGetLastError());
+  if (hipError_t _le = hipGetLastError(); _le != hipSuccess)
+    return hip_error(_le, "batched_copy_kernel launch");
   BB_HIP_TRY(hipEventRecord(g_copy_stage.ev, stream));
   return {};
 }
@@ -173,7 +174,8 @@ Result<void> fill_pattern(void* dev_ptr, uint64_t nbytes, uint64_t seed,
   if (nwords > 0) {
     fill_kernel<<<grid_for(nwords / 4), kBlock, 0, stream>>>(
         static_cast<ulong1*>(dev_ptr), nwords, seed);
-    BB_HIP_TRY(hipGetLastError());
+    if (hipError_t _le = hipGetLastError(); _le != hipSuccess)
+      return hip_error(_le, "fill_kernel launch");
   }
   return {};
 }
@@ -187,7 +189,8 @@ Result<uint64_t> verify_pattern(const void* dev_ptr, uint64_t nbytes, uint64_t s
   if (nwords > 0) {
     verify_kernel<<<grid_for(nwords / 4), kBlock, 0, stream>>>(
         static_cast<const ulong1*>(dev_ptr), nwords, seed, d_bad);
-    BB_HIP_TRY(hipGetLastError());
+    if (hipError_t _le = hipGetLastError(); _le != hipSuccess)
+      return hip_error(_le, "verify_kernel launch");
   }
   uint64_t bad = 0;
   BB_HIP_TRY(hipMemcpyAsync(&bad, d_bad, 8, hipMemcpyDeviceToHost, stream));
@@ -417,9 +420,11 @@ Result<void> fused_put(const PutDesc* descs, uint32_t n, uint64_t* out_digests,
   if (blocks < 1) blocks = 1;
   fused_put_kernel<<<static_cast<int>(blocks), kBlock, 0, stream>>>(
       d_segs, d_prefix, n, total, d_out);
-  BB_HIP_TRY(hipGetLastError());
+  if (hipError_t _le = hipGetLastError(); _le != hipSuccess)
+    return hip_error(_le, "fused_put_kernel launch");
   fused_put_finalize_kernel<<<(n + 255) / 256, 256, 0, stream>>>(d_segs, n, d_out);
-  BB_HIP_TRY(hipGetLastError());
+  if (hipError_t _le = hipGetLastError(); _le != hipSuccess)
+    return hip_error(_le, "fused_put_finalize_kernel launch");
   BB_HIP_TRY(hipMemcpyAsync(h_out, d_out, out_bytes, hipMemcpyDeviceToHost, stream));
   BB_HIP_TRY(hipStreamSynchronize(stream));
   std::memcpy(out_digests, h_out, out_bytes);
