@@ -68,12 +68,17 @@ class PoolMapper {
     if (dev_map_failed_.count(pool_key)) return nullptr;
     void* dp = nullptr;
     if (hipHostGetDevicePointer(&dp, base, 0) != hipSuccess) {
+      // the probe failing is EXPECTED for not-yet-registered memory — clear
+      // the sticky thread error so a later launch check doesn't inherit it
+      (void)hipGetLastError();
       if (hipHostRegister(base, size, hipHostRegisterDefault) != hipSuccess) {
+        (void)hipGetLastError();
         dev_map_failed_.insert(pool_key);
         return nullptr;
       }
       registered_.push_back(base);
       if (hipHostGetDevicePointer(&dp, base, 0) != hipSuccess) {
+        (void)hipGetLastError();
         dev_map_failed_.insert(pool_key);
         return nullptr;
       }
@@ -94,6 +99,7 @@ class PoolMapper {
     if (!from_hex(handle_hex, &h, sizeof(h))) return nullptr;
     void* p = nullptr;
     if (hipIpcOpenMemHandle(&p, h, hipIpcMemLazyEnablePeerAccess) != hipSuccess) {
+      (void)hipGetLastError();  // don't leak the probe error to launch checks
       ipc_failed_.insert(handle_hex);
       return nullptr;
     }
