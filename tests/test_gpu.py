@@ -626,6 +626,101 @@ class TestGpuDaemons:
                     p.kill()
 
 
+
+    def test_cross_process_shm_pool_fused(self, tmp_path):
+        """GPU client against a RAM_CPU pool owned by a SEPARATE worker
+        process: the pool is shm-mapped, GPU-mapped (hipHostRegister), and
+        the fused copy+digest kernel writes/reads it directly over PCIe.
+        Regression test: the first hipHostGetDevicePointer probe fails by
+        design and must not poison later kernel-launch error checks."""
+        import json as _json
+        import signal
+        import socket
+        import subprocess
+        repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        bin_dir = os.path.join(repo, "bin")
+        if not os.path.exists(os.path.join(bin_dir, "coordd")):
+            pytest.skip("daemons not built")
+        g = bb.core.gpu
+
+        def free_port():
+            s = socket.socket()
+            s.bind(("127.0.0.1", 0))
+            p = s.getsockname()[1]
+            s.close()
+            return p
+
+        cp, kp = free_port(), free_port()
+        procs = []
+
+        def spawn(args, logname):
+            f = open(tmp_path / logname, "w")
+            procs.append(subprocess.Popen(args, stdout=f,
+                                          stderr=subprocess.STDOUT))
+
+        try:
+            spawn([f"{bin_dir}/coordd", "--listen-host", "127.0.0.1",
+                   "--listen-port", str(cp)], "c.log")
+            time.sleep(0.3)
+            spawn([f"{bin_dir}/keystoned",
+                   "--listen-address", f"127.0.0.1:{kp}",
+                   "--coord-endpoint", f"127.0.0.1:{cp}",
+                   "--metrics-address", "127.0.0.1:0"], "k.log")
+            time.sleep(0.3)
+            cfg = {"worker_id": "shmw0",
+                   "coord_endpoint": f"127.0.0.1:{cp}",
+                   "data_listen_address": "127.0.0.1:0",
+                   "pools": [{"pool_id": "shm-pool",
+                              "storage_class": "RAM_CPU",
+                              "size_bytes": 256 << 20}]}
+            (tmp_path / "w.json").write_text(_json.dumps(cfg))
+            spawn([f"{bin_dir}/workerd", "--config",
+                   str(tmp_path / "w.json")], "w.log")
+
+            o = bb.ClientOptions()
+            o.keystone_endpoint = f"127.0.0.1:{kp}"
+            c = bb.Client(o)
+            deadline = time.time() + 15
+            while True:
+                try:
+                    c.connect()
+                    if c.memory_pools():
+                        break
+                except Exception:
+                    pass
+                assert time.time() < deadline, "cluster did not assemble"
+                time.sleep(0.3)
+
+            gcl = bb.GpuClient(c, 0)
+            gcl.init()
+            N, S = 8, 1 * MB
+            src = g.malloc(N * S)
+            dst = g.malloc(N * S)
+            g.fill_pattern(src, N * S, seed=3)
+            g.sync()
+            items = [("shx%d" % i, src + i * S, S) for i in range(N)]
+            cfgp = bb.PlacementConfig()
+            cfgp.checksum = True
+            assert gcl.batch_put_device(items, cfgp) == [0] * N
+            assert gcl.batch_get_device(
+                [(k, dst + i * S, S) for i, (k, _, _) in enumerate(items)],
+                verify=True) == [0] * N
+            g.sync()
+            assert g.verify_pattern(dst, N * S, seed=3) == 0
+            c.batch_remove([k for k, _, _ in items])
+            c.close()
+            g.free(src)
+            g.free(dst)
+        finally:
+            for p in procs:
+                p.send_signal(signal.SIGTERM)
+            for p in procs:
+                try:
+                    p.wait(timeout=5)
+                except subprocess.TimeoutExpired:
+                    p.kill()
+
+
 class TestGpuStriping:
     def test_striped_device_put_across_hbm_pools(self):
         """max_workers_per_copy=2 through GpuClient (the v1 multi-shard
