@@ -27,7 +27,10 @@ Error hip_err(hipError_t e, const char* what) {
 #define BB_HIP(expr)                                  \
   do {                                                \
     hipError_t _e = (expr);                           \
-    if (_e != hipSuccess) return hip_err(_e, #expr);  \
+    if (_e != hipSuccess) {                           \
+      (void)hipGetLastError(); /* consume sticky */   \
+      return hip_err(_e, #expr);                      \
+    }                                                 \
   } while (0)
 
 struct KeyMsg {

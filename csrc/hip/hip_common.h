@@ -17,7 +17,10 @@ inline Error hip_error(hipError_t e, const char* what) {
 #define BB_HIP_TRY(expr)                                        \
   do {                                                          \
     hipError_t _e = (expr);                                     \
-    if (_e != hipSuccess) return ::blackbird::gpu::hip_error(_e, #expr); \
+    if (_e != hipSuccess) {                                     \
+      (void)hipGetLastError(); /* consume sticky thread error */ \
+      return ::blackbird::gpu::hip_error(_e, #expr);            \
+    }                                                           \
   } while (0)
 
 }  // namespace blackbird::gpu
