@@ -930,6 +930,68 @@ class TestBatchSessions:
         finally:
             cl.stop()
 
+    def test_session_survives_worker_death(self):
+        """A session bound to placements on a worker that DIES must degrade
+        transparently: the keystone drops the dead worker's copies (epoch
+        bump invalidates the session + placement cache), and the next put
+        step re-places the whole batch on the surviving worker with zero
+        caller-visible errors."""
+        cl = Cluster(n_workers=2, pool_bytes=256 * MB,
+                     storage_class=bb.StorageClass.RAM_GPU)
+        g = bb.core.gpu
+        try:
+            c = cl.client()
+            gcl = bb.GpuClient(c, 0)
+            gcl.init()
+            gcl.set_placement_cache(True)
+            N, S = 8, 256 * 1024
+            src = g.malloc(N * S)
+            dst = g.malloc(N * S)
+            cfg = bb.PlacementConfig()
+            cfg.replace = True
+            cfg.preferred_worker = "w0"  # pin the session to the victim
+            pb = bb.make_put_batch([("wd%02d" % i, src + i * S, S)
+                                    for i in range(N)])
+            gb = bb.make_get_batch([("wd%02d" % i, dst + i * S, S)
+                                    for i in range(N)])
+            for step in range(3):  # establish + ride the session
+                blobs = [os.urandom(S) for _ in range(N)]
+                for i, b in enumerate(blobs):
+                    g.upload(src + i * S, b)
+                assert gcl.batch_put_prepared(pb, cfg)
+                assert gcl.batch_get_prepared(gb)
+            assert gcl.session_put_steps >= 1
+            ks = cl.keystone.service()
+            sh = ks.get_workers("wd00").copies[0].shards[0]
+            assert sh.worker_id == "w0"
+
+            cl.workers[0].stop()  # heartbeat TTL 1s → declared dead
+            deadline = time.time() + 10
+            while time.time() < deadline:
+                try:
+                    if ks.get_workers("wd00").copies[0].shards[0].worker_id \
+                            != "w0":
+                        break  # re-replicated to w1 already
+                except Exception:
+                    break  # or dropped entirely (no surviving replicas)
+                time.sleep(0.1)
+
+            cfg2 = bb.PlacementConfig()
+            cfg2.replace = True  # no preferred worker: place on survivors
+            blobs = [os.urandom(S) for _ in range(N)]
+            for i, b in enumerate(blobs):
+                g.upload(src + i * S, b)
+            assert gcl.batch_put_prepared(pb, cfg2)  # re-places, no error
+            assert gcl.batch_get_prepared(gb)
+            for i, b in enumerate(blobs):
+                assert g.download(dst + i * S, S) == b, i
+            assert ks.get_workers("wd00").copies[0].shards[0].worker_id == "w1"
+            c.close()
+            g.free(src)
+            g.free(dst)
+        finally:
+            cl.stop()
+
     def test_one_shot_token_commit(self):
         """Even without a reusable session (cache off, replace off), an
         all-fused batch put commits by token + digests — BATCH_PUT_COMPLETE
