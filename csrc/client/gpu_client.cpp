@@ -689,9 +689,18 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
     for (size_t j = 0; j < fused_hash_idx.size(); ++j) {
       const auto& it = items[fused_hash_idx[j]];
       if (fused_digests[j] == 0) continue;  // 0 marks "no digest"
-      placement_cache_[it.key] = {fused_hash_loc[j].first,
-                                  fused_hash_loc[j].second, it.size,
-                                  fused_digests[j]};
+      CachedPlacement np{fused_hash_loc[j].first, fused_hash_loc[j].second,
+                         it.size, fused_digests[j]};
+      auto [cit, inserted] = placement_cache_.try_emplace(it.key, np);
+      if (!inserted) {
+        // overwrite with a MOVED placement (e.g. re-placed after a worker
+        // death): sessions hold descs with the old pool addresses — they
+        // must die here, or a session get would read freed/stale memory
+        if (cit->second.pool_id != np.pool_id ||
+            cit->second.offset != np.offset || cit->second.size != np.size)
+          ++cache_epoch_;
+        cit->second = np;
+      }
     }
     if (placement_cache_.size() > (1u << 20)) {
       placement_cache_.clear();

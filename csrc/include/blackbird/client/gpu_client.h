@@ -61,6 +61,13 @@ class GpuClient {
   // full path. Sessions are bound to one item list: the item pointers are
   // re-validated each step. Requires replace=true, replication=1,
   // checksum=true, and the placement cache enabled.
+  // Worker death: a re-placing put bumps the cache epoch (moved placements
+  // kill sessions holding the old pool addresses). A get-session step
+  // BEFORE any re-put reads the dead pool one-sidedly: across processes
+  // (production: dmabuf-IPC imports pin the memory) that is a digest
+  // mismatch → RPC fallback; an EMBEDDED worker that frees its pool
+  // in-process can fault — stop in-process workers only after their
+  // clients.
   struct BatchPutSession {
     uint64_t token = 0;  // server put-session token (0 = not established)
     uint64_t cache_epoch = 0;
