@@ -11,6 +11,16 @@
 namespace blackbird {
 
 namespace {
+// relaxed atomic load of a hot ObjectMeta field for SHARED-lock readers
+// (session commits mutate these fields under the shared lock)
+template <typename T>
+inline T rload(const T& f) {
+  return std::atomic_ref<T>(const_cast<T&>(f)).load(std::memory_order_relaxed);
+}
+
+}  // namespace
+
+namespace {
 std::string random_id() {
   static std::mt19937_64 rng(std::random_device{}());
   char buf[20];
@@ -96,8 +106,10 @@ bool KeystoneService::is_leader() const {
 bool KeystoneService::object_exists(const ObjectKey& key) {
   std::shared_lock lk(objects_mu_);
   auto it = objects_.find(key);
-  return it != objects_.end() && it->second.state == ObjectState::COMMITTED &&
-         !it->second.expired(now_ms());
+  if (it == objects_.end()) return false;
+  const ObjectMeta& m = it->second;
+  if (rload(m.state) != ObjectState::COMMITTED) return false;
+  return !(m.ttl_ms > 0 && now_ms() > rload(m.created_ms) + m.ttl_ms);
 }
 
 Result<GetWorkersResponse> KeystoneService::get_workers(const ObjectKey& key) {
@@ -180,6 +192,22 @@ void apply_shard_digests(ObjectMeta& m, uint64_t checksum,
     }
   }
 }
+
+// shared-lock variant: digest slots may be read concurrently by get batches
+void apply_shard_digests_atomic(ObjectMeta& m, uint64_t checksum,
+                                const std::vector<std::vector<uint64_t>>& sd) {
+  for (size_t c = 0; c < m.copies.size(); ++c) {
+    auto& shards = m.copies[c].shards;
+    if (c < sd.size() && sd[c].size() == shards.size()) {
+      for (size_t s = 0; s < shards.size(); ++s)
+        std::atomic_ref<uint64_t>(shards[s].digest)
+            .store(sd[c][s], std::memory_order_relaxed);
+    } else if (shards.size() == 1) {
+      std::atomic_ref<uint64_t>(shards[0].digest)
+          .store(checksum, std::memory_order_relaxed);
+    }
+  }
+}
 }  // namespace
 
 Result<void> KeystoneService::put_complete(
@@ -237,7 +265,9 @@ std::vector<ObjectSummary> KeystoneService::list_objects(
   std::shared_lock lk(objects_mu_);
   for (const auto& [key, meta] : objects_) {
     if (out.size() >= limit) break;
-    if (meta.state != ObjectState::COMMITTED || meta.expired(now)) continue;
+    if (rload(meta.state) != ObjectState::COMMITTED ||
+        (meta.ttl_ms > 0 && now > rload(meta.created_ms) + meta.ttl_ms))
+      continue;
     if (!prefix.empty() && key.rfind(prefix, 0) != 0) continue;
     ObjectSummary s;
     s.key = key;
@@ -398,33 +428,44 @@ BatchPutStartResponse KeystoneService::batch_put_start(
 
 std::vector<int32_t> KeystoneService::batch_put_complete(
     const std::vector<PutCompleteRequest>& reqs) {
-  // one lock + one view bump for the whole batch (the per-key path costs a
-  // unique_lock and a bump each — measurable at 64 KiB object sizes)
+  // one SHARED lock + one view bump for the whole batch: commit batches
+  // from N ranks run concurrently (field mutations via atomic_ref — the
+  // same concurrency argument as commit_token; structural changes hold the
+  // lock exclusively and are therefore excluded)
   std::vector<int32_t> out(reqs.size(), 0);
   const uint64_t now = now_ms();
   bool any = false;
   {
-    std::unique_lock lk(objects_mu_);
+    std::shared_lock lk(objects_mu_);
     for (size_t i = 0; i < reqs.size(); ++i) {
       auto it = objects_.find(reqs[i].key);
       if (it == objects_.end()) {
         out[i] = static_cast<int32_t>(ErrorCode::OBJECT_NOT_FOUND);
         continue;
       }
-      if (it->second.state == ObjectState::COMMITTED) {
+      ObjectMeta& m = it->second;
+      const auto state = std::atomic_ref<ObjectState>(m.state)
+                             .load(std::memory_order_acquire);
+      if (state == ObjectState::COMMITTED) {
         // idempotent: a commit retried after a leader failover (the first
         // attempt applied + replicated, the reply was lost) carries the
         // same digest — report success, not INVALID_STATE
-        out[i] = it->second.checksum == reqs[i].checksum
+        out[i] = std::atomic_ref<uint64_t>(m.checksum)
+                             .load(std::memory_order_relaxed) ==
+                         reqs[i].checksum
                      ? 0
                      : static_cast<int32_t>(ErrorCode::INVALID_STATE);
         continue;
       }
-      it->second.state = ObjectState::COMMITTED;
-      it->second.checksum = reqs[i].checksum;
-      apply_shard_digests(it->second, reqs[i].checksum, reqs[i].shard_digests);
-      it->second.created_ms = now;
-      it->second.last_access_ms = now;
+      std::atomic_ref<uint64_t>(m.checksum)
+          .store(reqs[i].checksum, std::memory_order_relaxed);
+      apply_shard_digests_atomic(m, reqs[i].checksum, reqs[i].shard_digests);
+      std::atomic_ref<uint64_t>(m.created_ms)
+          .store(now, std::memory_order_relaxed);
+      std::atomic_ref<uint64_t>(m.last_access_ms)
+          .store(now, std::memory_order_relaxed);
+      std::atomic_ref<ObjectState>(m.state)
+          .store(ObjectState::COMMITTED, std::memory_order_release);
       mark_dirty_locked(reqs[i].key, false);
       any = true;
     }
@@ -884,10 +925,11 @@ std::vector<ShardPlacement> slice_shards(const std::vector<ShardPlacement>& srcs
 
 Result<void> KeystoneService::migrate_object(const ObjectKey& key,
                                              StorageClass target) {
-  // snapshot the object (no lock held during the transfer)
+  // snapshot the object (no lock held during the transfer). EXCLUSIVE so
+  // the full-struct copy cannot race shared-lock session commits.
   ObjectMeta snap;
   {
-    std::shared_lock lk(objects_mu_);
+    std::unique_lock lk(objects_mu_);
     auto it = objects_.find(key);
     if (it == objects_.end()) return Error{ErrorCode::OBJECT_NOT_FOUND, key};
     if (it->second.state != ObjectState::COMMITTED)
@@ -1025,12 +1067,13 @@ void KeystoneService::run_tiering_once() {
     {
       std::shared_lock lk(objects_mu_);
       for (const auto& [key, meta] : objects_) {
-        if (meta.state != ObjectState::COMMITTED || meta.copies.size() != 1)
+        if (rload(meta.state) != ObjectState::COMMITTED ||
+            meta.copies.size() != 1)
           continue;
         bool in_tier = !meta.copies[0].shards.empty();
         for (const auto& sh : meta.copies[0].shards)
           if (tier_rank(sh.storage_class) != rank) in_tier = false;
-        if (in_tier) cands.emplace_back(meta.last_access_ms, key);
+        if (in_tier) cands.emplace_back(rload(meta.last_access_ms), key);
       }
     }
     std::sort(cands.begin(), cands.end());
@@ -1080,13 +1123,14 @@ void KeystoneService::run_tiering_once() {
         {
           std::shared_lock lk(objects_mu_);
           for (const auto& [key, meta] : objects_) {
-            if (meta.state != ObjectState::COMMITTED ||
+            if (rload(meta.state) != ObjectState::COMMITTED ||
                 meta.copies.size() != 1 || meta.copies[0].shards.empty())
               continue;
-            if (meta.access_count < config_.promote_hot_threshold) continue;
+            const uint32_t ac = rload(meta.access_count);
+            if (ac < config_.promote_hot_threshold) continue;
             if (tier_rank(meta.copies[0].shards[0].storage_class) >
                 fastest->first)
-              hot.emplace_back(meta.access_count, key);
+              hot.emplace_back(ac, key);
           }
         }
         std::sort(hot.rbegin(), hot.rend());  // hottest first
@@ -1110,9 +1154,9 @@ void KeystoneService::run_tiering_once() {
 // ---------------------------------------------------------- failure repair
 
 Result<void> KeystoneService::repair_object(const ObjectKey& key) {
-  ObjectMeta snap;
+  ObjectMeta snap;  // exclusive: full-struct copy (see migrate_object)
   {
-    std::shared_lock lk(objects_mu_);
+    std::unique_lock lk(objects_mu_);
     auto it = objects_.find(key);
     if (it == objects_.end()) return Error{ErrorCode::OBJECT_NOT_FOUND, key};
     if (it->second.state != ObjectState::COMMITTED)
@@ -1209,8 +1253,8 @@ void KeystoneService::run_repair_once() {
   {
     std::shared_lock lk(objects_mu_);
     for (const auto& [key, meta] : objects_)
-      if (meta.state == ObjectState::COMMITTED && !meta.copies.empty() &&
-          meta.copies.size() < meta.replication)
+      if (rload(meta.state) == ObjectState::COMMITTED &&
+          !meta.copies.empty() && meta.copies.size() < meta.replication)
         degraded.push_back(key);
   }
   if (degraded.size() > config_.repair_max_per_cycle)
@@ -1265,7 +1309,9 @@ void KeystoneService::flush_dirty_now() {
   std::vector<coord::KV> puts;
   std::vector<std::string> dels;
   {
-    std::shared_lock lk(objects_mu_);
+    // EXCLUSIVE: serde reads every meta field; a concurrent shared-lock
+    // commit of the same key must not interleave with the serialization
+    std::unique_lock lk(objects_mu_);
     for (const auto& [key, removed] : batch) {
       if (removed) {
         dels.push_back(obj_prefix + key);
@@ -1306,7 +1352,7 @@ void KeystoneService::persist_loop() {
       }
       std::string blob;
       {
-        std::shared_lock lk(objects_mu_);
+        std::unique_lock lk(objects_mu_);  // see flush_dirty_now
         auto it = objects_.find(key);
         if (it == objects_.end()) continue;
         blob = serde::to_bytes(it->second);
@@ -1324,7 +1370,7 @@ void KeystoneService::persist_loop() {
     if (removed) {
       coord_->del(obj_prefix + key);
     } else {
-      std::shared_lock lk(objects_mu_);
+      std::unique_lock lk(objects_mu_);  // see flush_dirty_now
       auto it = objects_.find(key);
       if (it != objects_.end())
         coord_->put(obj_prefix + key, serde::to_bytes(it->second), 0);
@@ -1355,7 +1401,8 @@ Result<uint32_t> KeystoneService::compact_pool(const PoolId& pool_id,
   {
     std::shared_lock lk(objects_mu_);
     for (const auto& [key, meta] : objects_) {
-      if (meta.state != ObjectState::COMMITTED || meta.copies.size() != 1)
+      if (rload(meta.state) != ObjectState::COMMITTED ||
+          meta.copies.size() != 1)
         continue;
       for (const auto& sh : meta.copies[0].shards)
         if (sh.pool_id == pool_id) {
