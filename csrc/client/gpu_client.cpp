@@ -451,8 +451,9 @@ Result<void> GpuClient::session_kernel(Sess* sess, uint64_t* digests) {
 // session is not usable (caller runs the full path, which re-establishes it).
 std::optional<Result<std::vector<int32_t>>> GpuClient::try_session_put(
     const std::vector<DevPutItem>& items, BatchPutSession* sess) {
-  if (!sess || sess->token == 0 || sess->owner != this ||
-      sess->descs.size() != items.size() || items.empty())
+  const uint32_t dpi = sess ? std::max<uint32_t>(sess->descs_per_item, 1) : 1;
+  if (!sess || sess->token == 0 || sess->owner != this || items.empty() ||
+      sess->descs.size() != items.size() * dpi)
     return std::nullopt;
   {
     std::lock_guard<std::mutex> g(cache_mu_);
@@ -462,12 +463,14 @@ std::optional<Result<std::vector<int32_t>>> GpuClient::try_session_put(
     }
   }
   // the session is bound to one item list: same buffers, same order
+  // (dpi descs per item for replicated sessions — same src, one dst/copy)
   for (size_t i = 0; i < items.size(); ++i)
-    if (sess->descs[i].src != items[i].ptr ||
-        sess->descs[i].nbytes != items[i].size) {
-      sess->token = 0;
-      return std::nullopt;
-    }
+    for (uint32_t k = 0; k < dpi; ++k)
+      if (sess->descs[i * dpi + k].src != items[i].ptr ||
+          sess->descs[i * dpi + k].nbytes != items[i].size) {
+        sess->token = 0;
+        return std::nullopt;
+      }
   BB_TRACE_SCOPE("bb::session_put");
   // RPC 1 (8 bytes): flip the session's objects to PENDING — placements are
   // now pinned (tiering/eviction/repair only touch COMMITTED objects), so
@@ -479,14 +482,16 @@ std::optional<Result<std::vector<int32_t>>> GpuClient::try_session_put(
     sess->token = 0;  // stale/error before any write: clean fallback
     return std::nullopt;
   }
-  std::vector<uint64_t> digests(items.size(), 0);
+  std::vector<uint64_t> digests(sess->descs.size(), 0);
   auto rk = session_kernel(sess, digests.data());
   if (!rk.ok()) return {rk.error()};  // objects stay PENDING; GC reclaims
   serde::Enc e2;
   e2.num<uint64_t>(sess->token);
   e2.num<uint8_t>(0);  // keep the session for the next step
-  e2.num<uint32_t>(static_cast<uint32_t>(digests.size()));
-  for (uint64_t dg : digests) e2.num<uint64_t>(dg);
+  e2.num<uint32_t>(static_cast<uint32_t>(items.size()));
+  // one digest per ITEM (replicas hash identical bytes; take copy 0's)
+  for (size_t i = 0; i < items.size(); ++i)
+    e2.num<uint64_t>(digests[i * dpi]);
   auto r2 = c_.meta_call_raw(M::BATCH_COMMIT_TOKEN, e2.buf);
   if (!r2.ok()) {
     // placements changed mid-step (rare): fall back — the full path
@@ -501,7 +506,8 @@ std::optional<Result<std::vector<int32_t>>> GpuClient::try_session_put(
     std::lock_guard<std::mutex> g(cache_mu_);
     if (sess->cache_epoch == cache_epoch_)
       for (size_t j = 0; j < sess->entries.size(); ++j)
-        static_cast<CachedPlacement*>(sess->entries[j])->digest = digests[j];
+        static_cast<CachedPlacement*>(sess->entries[j])->digest =
+            digests[j * dpi];
   }
   session_put_steps_.fetch_add(1);
   return {Result<std::vector<int32_t>>(std::vector<int32_t>(items.size(), 0))};
@@ -512,14 +518,14 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
     BatchPutSession* sess) {
   if (auto fast = try_session_put(items, sess)) return std::move(*fast);
   BB_TRACE_SCOPE("bb::batch_put");
-  // establish: a reusable session for upsert steps (token kept server-side)
-  const bool establish = sess != nullptr && cfg.replace &&
-                         cfg.replication <= 1 && cfg.checksum &&
+  // establish: a reusable session for upsert steps (token kept server-side;
+  // replicated single-shard placements qualify — one desc per copy)
+  const bool establish = sess != nullptr && cfg.replace && cfg.checksum &&
                          placement_cache_on_ && fused_copy_;
   // one-shot tokens let ANY all-fused batch commit by token+digests instead
   // of re-sending every key in BATCH_PUT_COMPLETE (released at commit)
-  const bool want_token = establish ||
-                          (cfg.replication <= 1 && cfg.checksum && fused_copy_);
+  const bool want_token = establish || (cfg.checksum && fused_copy_);
+  const uint32_t dpi = std::max<uint32_t>(cfg.replication, 1);
   serde::Enc req;
   req.num<uint32_t>(static_cast<uint32_t>(items.size()));
   // uniform size when possible (the common batched pattern)
@@ -564,6 +570,66 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
     const uint8_t ncopies = d.num<uint8_t>();
     bool ok = true;
     bool hashed_in_fuse = false;
+    // fused copy+digest fast path: every copy base-resolvable + aligned
+    // (one desc per copy, digests identical across replicas)
+    const uint8_t* isrc = static_cast<const uint8_t*>(items[i].ptr);
+    bool all_fusable = fused_copy_ && cfg.checksum && ncopies == dpi &&
+                       (reinterpret_cast<uintptr_t>(isrc) & 15) == 0;
+    std::pair<uint16_t, uint64_t> places[8];
+    if (ncopies <= 8) {
+      // decode ALL copies first (the wire cursor must always advance),
+      // then decide fused vs general
+      for (uint8_t c = 0; c < ncopies; ++c)
+        places[c] = {d.num<uint16_t>(), d.num<uint64_t>()};
+      for (uint8_t c = 0; c < ncopies && all_fusable; ++c) {
+        PoolRef* pr = places[c].first < npools ? &pools[places[c].first]
+                                               : nullptr;
+        if (!pr || !pr->base ||
+            (reinterpret_cast<uintptr_t>(pr->base + places[c].second) & 15))
+          all_fusable = false;
+      }
+      if (all_fusable) {
+        for (uint8_t c = 0; c < ncopies; ++c) {
+          PoolRef& pr = pools[places[c].first];
+          fused_hash.push_back(
+              {isrc, pr.base + places[c].second, items[i].size});
+          if (c == 0)  // the verified-get cache reads copy 0
+            fused_hash_loc.emplace_back(pr.pool_id, places[c].second);
+        }
+        hashed_in_fuse = true;
+      } else {
+        // fall through: route the already-decoded copies the general way
+        for (uint8_t c = 0; c < ncopies && ok; ++c) {
+          auto [pi, off] = places[c];
+          if (pi >= npools) { ok = false; break; }
+          PoolRef& pr = pools[pi];
+          if (pr.base) {
+            if (fused_copy_)
+              fused.push_back({isrc, pr.base + off, items[i].size});
+            else {
+              hipError_t e = hipMemcpyAsync(pr.base + off, isrc,
+                                            items[i].size,
+                                            hipMemcpyDeviceToDevice,
+                                            streams_[si % kStreams]);
+              if (e != hipSuccess) {
+                (void)hipGetLastError();
+                ok = false;
+                break;
+              }
+              ++si;
+            }
+          } else {
+            ShardPlacement sp;
+            sp.pool_id = pr.pool_id;
+            sp.offset = off;
+            sp.length = items[i].size;
+            sp.access = pr.access;
+            staged_put_work.emplace_back(std::move(sp), isrc);
+            staged_put_idx.push_back(static_cast<uint32_t>(i));
+          }
+        }
+      }
+    } else
     for (uint8_t c = 0; c < ncopies; ++c) {
       const uint16_t pi = d.num<uint16_t>();
       const uint64_t off = d.num<uint64_t>();
@@ -572,13 +638,7 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
       const uint8_t* src = static_cast<const uint8_t*>(items[i].ptr);
       if (pr.base) {
         uint8_t* dst = pr.base + off;
-        const auto su = reinterpret_cast<uintptr_t>(src);
-        if (fused_copy_ && cfg.checksum && ncopies == 1 &&
-            ((su | reinterpret_cast<uintptr_t>(dst)) & 15) == 0) {
-          fused_hash.push_back({src, dst, items[i].size});
-          fused_hash_loc.emplace_back(pr.pool_id, off);
-          hashed_in_fuse = true;
-        } else if (fused_copy_) {
+        if (fused_copy_) {
           fused.push_back({src, dst, items[i].size});
         } else {
           hipError_t e = hipMemcpyAsync(dst, src, items[i].size,
@@ -661,8 +721,9 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
     serde::Enc e2;
     e2.num<uint64_t>(token);
     e2.num<uint8_t>(establish ? 0 : 1);  // one-shot: release after commit
-    e2.num<uint32_t>(static_cast<uint32_t>(fused_digests.size()));
-    for (uint64_t dg : fused_digests) e2.num<uint64_t>(dg);
+    e2.num<uint32_t>(static_cast<uint32_t>(fused_hash_idx.size()));
+    for (size_t j = 0; j < fused_hash_idx.size(); ++j)
+      e2.num<uint64_t>(fused_digests[j * dpi]);
     auto r2 = c_.meta_call_raw(M::BATCH_COMMIT_TOKEN, e2.buf);
     committed_by_token = r2.ok();
   }
@@ -674,8 +735,8 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
           PutCompleteRequest{items[committed_idx[j]].key, digests[j], {}});
     }
     for (size_t j = 0; j < fused_hash_idx.size(); ++j)
-      completes.reqs.push_back(
-          PutCompleteRequest{items[fused_hash_idx[j]].key, fused_digests[j], {}});
+      completes.reqs.push_back(PutCompleteRequest{
+          items[fused_hash_idx[j]].key, fused_digests[j * dpi], {}});
     if (!completes.reqs.empty()) {
       auto r = c_.meta_call<PutCompleteListMsg, StatusListMsg>(
           M::BATCH_PUT_COMPLETE, completes);
@@ -688,9 +749,9 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
     std::lock_guard<std::mutex> g(cache_mu_);
     for (size_t j = 0; j < fused_hash_idx.size(); ++j) {
       const auto& it = items[fused_hash_idx[j]];
-      if (fused_digests[j] == 0) continue;  // 0 marks "no digest"
+      if (fused_digests[j * dpi] == 0) continue;  // 0 marks "no digest"
       CachedPlacement np{fused_hash_loc[j].first, fused_hash_loc[j].second,
-                         it.size, fused_digests[j]};
+                         it.size, fused_digests[j * dpi]};
       auto [cit, inserted] = placement_cache_.try_emplace(it.key, np);
       if (!inserted) {
         // overwrite with a MOVED placement (e.g. re-placed after a worker
@@ -712,6 +773,7 @@ Result<std::vector<int32_t>> GpuClient::batch_put_device_v2(
         fused_hash_idx.size() == items.size()) {
       bool all_cached = true;
       sess->descs = fused_hash;
+      sess->descs_per_item = dpi;
       sess->plan.reset();  // descs changed: a kept plan would replay stale
       sess->plan_failed = false;
       sess->entries.clear();

@@ -74,6 +74,9 @@ class GpuClient {
     void* owner = nullptr;  // GpuClient that built it (entries point into
                             // that client's placement cache)
     std::vector<gpu::PutDesc> descs;  // src=user buffer, dst=pool range
+    // replicated sessions carry one desc PER COPY per item (same src, one
+    // dst per replica); descs.size() == items.size() * descs_per_item
+    uint32_t descs_per_item = 1;
     std::vector<void*> entries;       // CachedPlacement* digest slots
     // hipGraph replay of the step (descs fixed ⇒ captured once, one graph
     // launch per step); built lazily on the first session step

@@ -488,12 +488,13 @@ uint64_t KeystoneService::create_put_session(
     std::shared_lock lk(objects_mu_);
     for (const auto& r : reqs) {
       auto it = objects_.find(r.key);
-      // sessions only cover single-copy single-shard objects: the token
-      // commit path records ONE digest per object, and the client's fused
-      // write path resolves exactly one destination
-      if (it == objects_.end() || it->second.copies.size() != 1 ||
-          it->second.copies[0].shards.size() != 1)
+      // sessions cover single-SHARD copies (any replica count): the token
+      // commit records ONE digest per object — every replica holds the
+      // same bytes — and the client resolves one destination per copy
+      if (it == objects_.end() || it->second.copies.empty())
         return 0;
+      for (const auto& c : it->second.copies)
+        if (c.shards.size() != 1) return 0;
       s->metas.push_back(&it->second);
       s->sizes.push_back(it->second.size);
     }
@@ -581,9 +582,10 @@ Result<void> KeystoneService::commit_token(
     ObjectMeta* m = s->metas[i];
     std::atomic_ref<uint64_t>(m->checksum)
         .store(digests[i], std::memory_order_relaxed);
-    // sessions cover single-copy single-shard objects by construction
-    std::atomic_ref<uint64_t>(m->copies[0].shards[0].digest)
-        .store(digests[i], std::memory_order_relaxed);
+    // single-shard copies by construction; every replica gets the digest
+    for (auto& c : m->copies)
+      std::atomic_ref<uint64_t>(c.shards[0].digest)
+          .store(digests[i], std::memory_order_relaxed);
     std::atomic_ref<uint64_t>(m->created_ms)
         .store(now, std::memory_order_relaxed);  // TTL restarts at commit
     std::atomic_ref<uint64_t>(m->last_access_ms)

@@ -930,6 +930,68 @@ class TestBatchSessions:
         finally:
             cl.stop()
 
+    def test_replicated_session_fast_path(self):
+        """replication=2 batches ride the session fast path too: one desc
+        per copy (same source), both replicas rewritten in place each step,
+        the shared digest stamped on both shards. Kill one worker at the
+        end: the surviving replica still serves verified gets."""
+        cl = Cluster(n_workers=2, pool_bytes=256 * MB,
+                     storage_class=bb.StorageClass.RAM_GPU)
+        g = bb.core.gpu
+        try:
+            c = cl.client()
+            gcl = bb.GpuClient(c, 0)
+            gcl.init()
+            gcl.set_placement_cache(True)
+            N, S = 8, 256 * 1024
+            src = g.malloc(N * S)
+            dst = g.malloc(N * S)
+            cfg = bb.PlacementConfig()
+            cfg.replace = True
+            cfg.replication = 2
+            pb = bb.make_put_batch([("rs%02d" % i, src + i * S, S)
+                                    for i in range(N)])
+            gb = bb.make_get_batch([("rs%02d" % i, dst + i * S, S)
+                                    for i in range(N)])
+            ks = cl.keystone.service()
+            for step in range(4):
+                blobs = [os.urandom(S) for _ in range(N)]
+                for i, b in enumerate(blobs):
+                    g.upload(src + i * S, b)
+                assert gcl.batch_put_prepared(pb, cfg), step
+                assert gcl.batch_get_prepared(gb), step
+                for i, b in enumerate(blobs):
+                    assert g.download(dst + i * S, S) == b, (step, i)
+            assert gcl.session_put_steps >= 2, gcl.session_put_steps
+            info = ks.get_workers("rs03")
+            assert len(info.copies) == 2
+            assert info.checksum == g.checksum_cpu(blobs[3])
+            workers = {cp.shards[0].worker_id for cp in info.copies}
+            assert workers == {"w0", "w1"}
+            for cp in info.copies:
+                assert cp.shards[0].digest == info.checksum
+            # replica failover: stop the worker holding copy 0, verified
+            # gets still succeed from the surviving copy
+            victim = info.copies[0].shards[0].worker_id
+            cl.workers[int(victim[1])].stop()
+            deadline = time.time() + 10
+            while time.time() < deadline:
+                cps = ks.get_workers("rs03").copies
+                if all(cp.shards[0].worker_id != victim for cp in cps):
+                    break
+                time.sleep(0.1)
+            st = gcl.batch_get_device(
+                [("rs%02d" % i, dst + i * S, S) for i in range(N)],
+                verify=True)
+            assert st == [0] * N
+            for i, b in enumerate(blobs):
+                assert g.download(dst + i * S, S) == b, i
+            c.close()
+            g.free(src)
+            g.free(dst)
+        finally:
+            cl.stop()
+
     def test_session_survives_worker_death(self):
         """A session bound to placements on a worker that DIES must degrade
         transparently: the keystone drops the dead worker's copies (epoch

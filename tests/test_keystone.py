@@ -488,15 +488,24 @@ class TestPutSessions:
         with pytest.raises(Exception, match="SESSION_STALE"):
             ks.commit_token(tok, [0] * 4)
 
-    def test_session_not_granted_for_replicated(self, ks):
-        # token commits record ONE digest per object: multi-copy objects are
-        # excluded at session creation
+    def test_session_for_replicated_single_shard(self, ks):
+        """Replicated single-shard objects get sessions too (one desc per
+        copy client-side; the token commit stamps the shared digest on every
+        replica) — only STRIPED copies are excluded."""
         ks.register_pool(make_pool("p1", worker="w1"))
         cfg = bb.PlacementConfig()
         cfg.replication = 2
         ks.put_start("rep", 4096, cfg)
         ks.put_complete("rep", checksum=5)
-        assert ks.create_put_session(["rep"], 4096, cfg) == 0
+        tok = ks.create_put_session(["rep"], 4096, cfg)
+        assert tok != 0
+        ks.upsert_start_token(tok)
+        ks.commit_token(tok, [77])
+        info = ks.get_workers("rep")
+        assert info.checksum == 77
+        assert len(info.copies) == 2
+        for c in info.copies:
+            assert c.shards[0].digest == 77
 
     def test_session_not_granted_for_missing(self, ks):
         assert ks.create_put_session(["nope"], 4096, bb.PlacementConfig()) == 0
