@@ -241,19 +241,23 @@ def main():
     else:
         import numpy as np
         rng = np.random.default_rng(1234 + RANK)
-        blobs = [rng.integers(0, 256, size=S, dtype=np.uint8).tobytes()
-                 for _ in range(B)]
-        put_items = [(f"r{RANK}o{i}", blobs[i]) for i in range(B)]
+        # stable buffers (numpy arrays) so the host batch session binds them
+        arrs = [rng.integers(0, 256, size=S, dtype=np.uint8) for _ in range(B)]
+        put_items = [(f"r{RANK}o{i}", arrs[i]) for i in range(B)]
         keys = [k for k, _ in put_items]
+        no_replace = os.environ.get("BB_BENCH_NO_REPLACE") == "1"
+        cfg.replace = not no_replace
+        host_sess = bb.HostPutSession()
 
         def do_step(lane=0):
-            st = client.batch_put(put_items, cfg)
+            st = client.batch_put_session(put_items, cfg, host_sess)
             assert all(s == 0 for s in st), f"put failures: {st[:5]}"
             t0 = time.perf_counter()
             res = client.batch_get(keys)
             get_ms = (time.perf_counter() - t0) * 1e3
             assert all(s == 0 for s, _ in res)
-            client.batch_remove(keys)
+            if no_replace:
+                client.batch_remove(keys)
             return get_ms
         lanes = 1  # host tier: single lane
 
@@ -310,6 +314,31 @@ def main():
     # are visible in the official record. ----
     cold_gbps = None
     cold_steps = max(2, args.steps // 4)
+    if not use_gpu:
+        # host-tier cold regime: fresh keys, full control plane, removes
+        cold_cfg = bb.PlacementConfig()
+        for attr in ("replication", "checksum", "preferred_class",
+                     "preferred_worker"):
+            setattr(cold_cfg, attr, getattr(cfg, attr))
+        cold_cfg.replace = False
+        citems = [(f"r{RANK}cold{i}", arrs[i]) for i in range(B)]
+        ckeys = [k for k, _ in citems]
+
+        def cold_step():
+            st = client.batch_put(citems, cold_cfg)
+            assert all(s == 0 for s in st), "cold put"
+            res = client.batch_get(ckeys)
+            assert all(s == 0 for s, _ in res)
+            client.batch_remove(ckeys)
+
+        cold_step()  # warmup
+        barrier(dist)
+        t0c = time.perf_counter()
+        for _ in range(cold_steps):
+            cold_step()
+        barrier(dist)
+        elapsed_c = max_over_ranks(dist, time.perf_counter() - t0c)
+        cold_gbps = 2.0 * B * S * cold_steps * max(WORLD, 1) / elapsed_c / 1e9
     if use_gpu:
         cold_cfg = bb.PlacementConfig()
         for attr in ("replication", "checksum", "preferred_class",

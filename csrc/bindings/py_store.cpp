@@ -313,6 +313,11 @@ void bind_store(py::module_& m) {
       .def_readwrite("rpc_timeout_ms", &ClientOptions::rpc_timeout_ms)
       .def_readwrite("force_tcp", &ClientOptions::force_tcp);
 
+  py::class_<Client::HostPutSession>(m, "HostPutSession")
+      .def(py::init<>())
+      .def_property_readonly(
+          "active", [](const Client::HostPutSession& s) { return s.token != 0; });
+
   py::class_<Client>(m, "Client")
       .def(py::init<ClientOptions>(), py::arg("options") = ClientOptions{})
       .def("connect", [](Client& c) { unwrap_void(c.connect()); },
@@ -358,6 +363,26 @@ void bind_store(py::module_& m) {
         py::gil_scoped_release rel;
         return unwrap(c.batch_put(its, cfg));
       }, py::arg("items"), py::arg("config") = PlacementConfig{})
+      .def("batch_put_session",
+           [](Client& c,
+              const std::vector<std::pair<std::string, py::buffer>>& items,
+              const PlacementConfig& cfg, Client::HostPutSession& sess) {
+             std::vector<Client::PutItem> its;
+             std::vector<py::buffer_info> infos;
+             its.reserve(items.size());
+             infos.reserve(items.size());
+             for (auto& [k, b] : items) {
+               infos.push_back(const_cast<py::buffer&>(b).request());
+               its.push_back({k, infos.back().ptr,
+                              static_cast<uint64_t>(infos.back().size *
+                                                    infos.back().itemsize)});
+             }
+             py::gil_scoped_release rel;
+             return unwrap(c.batch_put(its, cfg, &sess));
+           },
+           py::arg("items"), py::arg("config"), py::arg("session"))
+      .def_property_readonly("host_session_steps",
+                             &Client::host_session_steps)
       .def("batch_get", [](Client& c, const std::vector<std::string>& keys) {
         std::vector<std::pair<int32_t, std::string>> res;
         {

@@ -564,9 +564,11 @@ bool Client::failover_retriable(int32_t st) {
 }
 
 Result<std::vector<int32_t>> Client::batch_put(const std::vector<PutItem>& items,
-                                               const PlacementConfig& cfg) {
+                                               const PlacementConfig& cfg,
+                                               HostPutSession* sess) {
+  if (auto fast = try_host_session_put(items, sess)) return std::move(*fast);
   const uint64_t gen = reconnect_gen_.load();
-  auto st = batch_put_once(items, cfg);
+  auto st = batch_put_once(items, cfg, sess);
   if (!st.ok()) return st;
   if (reconnect_gen_.load() == gen) return st;  // no failover: statuses final
   // a leader change happened during the batch: redo the items whose failure
@@ -579,11 +581,68 @@ Result<std::vector<int32_t>> Client::batch_put(const std::vector<PutItem>& items
       redo_idx.push_back(i);
     }
   if (redo.empty()) return st;
-  auto st2 = batch_put_once(redo, cfg);
+  auto st2 = batch_put_once(redo, cfg, nullptr);
   if (!st2.ok()) return st;  // keep the first answer
   for (size_t j = 0; j < redo_idx.size(); ++j)
     st.value()[redo_idx[j]] = st2.value()[j];
   return st;
+}
+
+// Token fast path for the host tier: placements unchanged since the last
+// step ⇒ two tiny RPCs around direct memcpys + CPU digests. Mirrors
+// GpuClient::try_session_put, including the ordering guarantee: the server
+// flips the session's objects to PENDING (pinning the placements) BEFORE
+// any byte is written.
+std::optional<std::vector<int32_t>> Client::try_host_session_put(
+    const std::vector<PutItem>& items, HostPutSession* sess) {
+  if (!sess || sess->token == 0 || sess->owner != this || items.empty())
+    return std::nullopt;
+  const uint32_t dpi = std::max<uint32_t>(sess->descs_per_item, 1);
+  if (sess->srcs.size() != items.size() ||
+      sess->dsts.size() != items.size() * dpi)
+    return std::nullopt;
+  for (size_t i = 0; i < items.size(); ++i)
+    if (sess->srcs[i] != items[i].data || sess->sizes[i] != items[i].size) {
+      sess->token = 0;
+      return std::nullopt;
+    }
+  serde::Enc e1;
+  e1.num<uint64_t>(sess->token);
+  auto r1 = meta_call_raw(M::BATCH_UPSERT_START, e1.buf, opts_.rpc_timeout_ms);
+  if (!r1.ok()) {
+    sess->token = 0;  // stale before any write: clean fallback
+    return std::nullopt;
+  }
+  std::vector<uint64_t> digests(items.size(), 0);
+  {
+    std::atomic<size_t> next{0};
+    const int nthreads = std::max(
+        1, std::min<int>(opts_.io_threads, static_cast<int>(items.size())));
+    std::vector<std::future<void>> futs;
+    for (int t = 0; t < nthreads; ++t)
+      futs.push_back(std::async(std::launch::async, [&] {
+        for (size_t i = next.fetch_add(1); i < items.size();
+             i = next.fetch_add(1)) {
+          for (uint32_t k = 0; k < dpi; ++k)
+            std::memcpy(sess->dsts[i * dpi + k], items[i].data,
+                        items[i].size);
+          digests[i] = gpu::checksum_cpu(items[i].data, items[i].size);
+        }
+      }));
+    for (auto& f : futs) f.get();
+  }
+  serde::Enc e2;
+  e2.num<uint64_t>(sess->token);
+  e2.num<uint8_t>(0);  // keep the session
+  e2.num<uint32_t>(static_cast<uint32_t>(digests.size()));
+  for (uint64_t dg : digests) e2.num<uint64_t>(dg);
+  auto r2 = meta_call_raw(M::BATCH_COMMIT_TOKEN, e2.buf, opts_.rpc_timeout_ms);
+  if (!r2.ok()) {
+    sess->token = 0;  // placements changed mid-step: full path re-places
+    return std::nullopt;
+  }
+  host_session_steps_.fetch_add(1);
+  return std::vector<int32_t>(items.size(), 0);
 }
 
 uint8_t* Client::host_pool_base(const PoolId& id, AccessInfo* access) {
@@ -614,7 +673,10 @@ struct HostPoolRef {
 // on the fly, commit by one-shot token when the whole batch qualifies —
 // BATCH_PUT_COMPLETE never re-sends keys on the common path.
 Result<std::vector<int32_t>> Client::batch_put_once_v2(
-    const std::vector<PutItem>& items, const PlacementConfig& cfg) {
+    const std::vector<PutItem>& items, const PlacementConfig& cfg,
+    HostPutSession* sess) {
+  const bool establish =
+      sess != nullptr && cfg.replace && cfg.checksum && !opts_.force_tcp;
   serde::Enc req;
   req.num<uint32_t>(static_cast<uint32_t>(items.size()));
   uint64_t uniform = items.empty() ? 1 : items[0].size;
@@ -625,7 +687,7 @@ Result<std::vector<int32_t>> Client::batch_put_once_v2(
     for (auto& it : items) req.num<uint64_t>(it.size);
   for (auto& it : items) req.str(it.key);
   serde::put(req, cfg);
-  const bool want_token = cfg.replication <= 1 && cfg.checksum;
+  const bool want_token = establish || cfg.checksum;
   req.num<uint8_t>(want_token ? 1 : 0);
   auto resp = meta_call_raw(M::BATCH_PUT_START2, req.buf, opts_.rpc_timeout_ms);
   if (!resp.ok()) return resp.error();
@@ -660,6 +722,28 @@ Result<std::vector<int32_t>> Client::batch_put_once_v2(
     }
   }
   if (!d.ok()) return Error{ErrorCode::PROTOCOL_ERROR, "bad v2 response"};
+
+  // establishment bookkeeping: one dst per copy per item, all host-mapped
+  const uint32_t dpi = std::max<uint32_t>(cfg.replication, 1);
+  std::vector<uint8_t*> sess_dsts;
+  bool all_mapped = establish;
+  if (establish) {
+    sess_dsts.assign(items.size() * dpi, nullptr);
+    for (size_t i = 0; i < items.size() && all_mapped; ++i) {
+      if (statuses[i] != 0 || placed[i].size() != dpi) {
+        all_mapped = false;
+        break;
+      }
+      for (uint32_t k = 0; k < dpi; ++k) {
+        HostPoolRef& pr = pools[placed[i][k].first];
+        if (!pr.base) {
+          all_mapped = false;
+          break;
+        }
+        sess_dsts[i * dpi + k] = pr.base + placed[i][k].second;
+      }
+    }
+  }
 
   // transfers + digests fan out per item (digest runs while the bytes are
   // still cache-hot from the memcpy)
@@ -701,15 +785,32 @@ Result<std::vector<int32_t>> Client::batch_put_once_v2(
   bool all_ok = token != 0;
   for (auto st : statuses)
     if (st != 0) { all_ok = false; break; }
+  const bool keep_session = all_ok && establish && all_mapped;
   bool committed = false;
   if (all_ok) {
     serde::Enc e2;
     e2.num<uint64_t>(token);
-    e2.num<uint8_t>(1);  // one-shot: release after commit
+    e2.num<uint8_t>(keep_session ? 0 : 1);  // one-shot unless establishing
     e2.num<uint32_t>(static_cast<uint32_t>(digests.size()));
     for (uint64_t dg : digests) e2.num<uint64_t>(dg);
     committed =
         meta_call_raw(M::BATCH_COMMIT_TOKEN, e2.buf, opts_.rpc_timeout_ms).ok();
+  }
+  if (keep_session && committed) {
+    sess->token = token;
+    sess->descs_per_item = dpi;
+    sess->dsts = std::move(sess_dsts);
+    sess->srcs.clear();
+    sess->sizes.clear();
+    sess->srcs.reserve(items.size());
+    sess->sizes.reserve(items.size());
+    for (const auto& it : items) {
+      sess->srcs.push_back(it.data);
+      sess->sizes.push_back(it.size);
+    }
+    sess->owner = this;
+  } else if (sess) {
+    sess->token = 0;
   }
   if (!committed) {
     PutCompleteListMsg completes;
@@ -827,9 +928,10 @@ Result<std::vector<std::pair<int32_t, std::string>>> Client::batch_get_once_v2(
 }
 
 Result<std::vector<int32_t>> Client::batch_put_once(const std::vector<PutItem>& items,
-                                               const PlacementConfig& cfg) {
+                                               const PlacementConfig& cfg,
+                                               HostPutSession* sess) {
   if (!opts_.force_tcp && cfg.max_workers_per_copy <= 1 && !items.empty())
-    return batch_put_once_v2(items, cfg);
+    return batch_put_once_v2(items, cfg, sess);
   BatchPutStartRequest breq;
   breq.requests.reserve(items.size());
   for (const auto& it : items)

@@ -11,10 +11,12 @@
 // over an IO thread pool; gets fail over across replicas.
 #pragma once
 
+#include <atomic>
 #include <functional>
 #include <map>
 #include <memory>
 #include <mutex>
+#include <optional>
 #include <string>
 #include <vector>
 
@@ -69,19 +71,36 @@ class Client {
     const void* data;
     uint64_t size;
   };
+  // Host-tier batch session (the DRAM twin of GpuClient's BatchPutSession):
+  // callers re-putting the SAME batch (same keys/buffers/sizes, replace
+  // mode) pay two tiny RPCs per step — upsert start, token+digest commit —
+  // around direct memcpys into the mapped pool ranges. Established by the
+  // first full batch_put when every copy resolved to a mapped host pool;
+  // any server-side placement change invalidates it transparently.
+  struct HostPutSession {
+    uint64_t token = 0;
+    uint32_t descs_per_item = 1;       // one dst per copy
+    std::vector<uint8_t*> dsts;        // items.size() * descs_per_item
+    std::vector<const void*> srcs;     // bound item buffers
+    std::vector<uint64_t> sizes;
+    void* owner = nullptr;
+  };
   // One metadata RPC for the whole batch; per-item status out. A leader
   // failover mid-batch (put_start answered by the old leader, put_complete
   // by the new one that never saw the PENDING objects) is resumed
   // transparently: items that failed with failover-shaped errors WHILE a
   // reconnect happened are redone once against the new leader.
   Result<std::vector<int32_t>> batch_put(const std::vector<PutItem>& items,
-                                         const PlacementConfig& cfg = {});
+                                         const PlacementConfig& cfg = {},
+                                         HostPutSession* sess = nullptr);
   Result<std::vector<std::pair<int32_t, std::string>>> batch_get(
       const std::vector<ObjectKey>& keys);
   // bumped every time meta_call_raw re-established the leader connection
   uint64_t reconnect_generation() const { return reconnect_gen_.load(); }
   // status looks like a lost-leader symptom (worth one redo after failover)?
   static bool failover_retriable(int32_t st);
+  // steps served by the host session fast path (tests/bench introspection)
+  uint64_t host_session_steps() const { return host_session_steps_.load(); }
 
   // ------------------------------------------------------ cluster view
   Result<ClusterStats> cluster_stats();
@@ -134,8 +153,13 @@ class Client {
   std::map<PoolId, AccessInfo> pool_cache_;
   std::mutex reconnect_mu_;  // one thread rediscovers/reconnects at a time
   std::atomic<uint64_t> reconnect_gen_{0};
+  std::atomic<uint64_t> host_session_steps_{0};
   Result<std::vector<int32_t>> batch_put_once(const std::vector<PutItem>& items,
-                                              const PlacementConfig& cfg);
+                                              const PlacementConfig& cfg,
+                                              HostPutSession* sess);
+  // token fast path; nullopt -> run the full path (which re-establishes)
+  std::optional<std::vector<int32_t>> try_host_session_put(
+      const std::vector<PutItem>& items, HostPutSession* sess);
   Result<void> put_once(const ObjectKey& key, const void* data, uint64_t size,
                         const PlacementConfig& cfg);
   Result<std::vector<std::pair<int32_t, std::string>>> batch_get_once(
@@ -144,7 +168,8 @@ class Client {
   // token commits) — the host-tier twin of GpuClient's v2 paths. Single-shard
   // placements only; *fallback signals "run the v1 path instead".
   Result<std::vector<int32_t>> batch_put_once_v2(
-      const std::vector<PutItem>& items, const PlacementConfig& cfg);
+      const std::vector<PutItem>& items, const PlacementConfig& cfg,
+      HostPutSession* sess);
   Result<std::vector<std::pair<int32_t, std::string>>> batch_get_once_v2(
       const std::vector<ObjectKey>& keys, bool* fallback);
   // host-visible base of a pool (same-process host pool or SHM mapping);

@@ -525,3 +525,72 @@ class TestDiscovery:
         c.put("via-discovery", b"found you")
         assert c.get("via-discovery") == b"found you"
         c.close()
+
+
+class TestHostSessions:
+    def test_host_session_fast_path(self):
+        """The DRAM-tier twin of the GPU batch sessions: after the first
+        full batch_put (replace mode, stable buffers), steps ride the token
+        fast path — two tiny RPCs around direct memcpys. Data and digests
+        stay correct as the buffer CONTENTS change between steps, and
+        server-side interference invalidates the session transparently."""
+        import numpy as np
+        cl = Cluster(n_workers=1, pool_bytes=64 << 20)
+        try:
+            c = cl.client()
+            N, S = 16, 8192
+            arrs = [np.zeros(S, np.uint8) for _ in range(N)]
+            items = [("hs%02d" % i, arrs[i]) for i in range(N)]
+            keys = [k for k, _ in items]
+            cfg = bb.PlacementConfig()
+            cfg.replace = True
+            cfg.checksum = True
+            sess = bb.HostPutSession()
+            ks = cl.keystone.service()
+            for step in range(4):
+                blobs = [os.urandom(S) for _ in range(N)]
+                for a, b in zip(arrs, blobs):
+                    a[:] = np.frombuffer(b, np.uint8)
+                st = c.batch_put_session(items, cfg, sess)
+                assert st == [0] * N, (step, st[:4])
+                res = c.batch_get(keys)
+                for i, (s_, got) in enumerate(res):
+                    assert s_ == 0 and got == blobs[i], (step, i)
+                info = ks.get_workers(keys[5])
+                assert info.checksum == bb.core.gpu.checksum_cpu(blobs[5])
+            assert sess.active
+            assert c.host_session_steps >= 3, c.host_session_steps
+            # interference: any placement change bumps the epoch → the next
+            # step falls back to the full path, then re-establishes
+            ks.put_start("intruder", 4096, bb.PlacementConfig())
+            ks.put_complete("intruder", checksum=1)
+            ks.remove_object("intruder")
+            fast_before = c.host_session_steps
+            blobs = [os.urandom(S) for _ in range(N)]
+            for a, b in zip(arrs, blobs):
+                a[:] = np.frombuffer(b, np.uint8)
+            assert c.batch_put_session(items, cfg, sess) == [0] * N
+            assert c.host_session_steps == fast_before  # full path this step
+            assert c.batch_put_session(items, cfg, sess) == [0] * N
+            assert c.host_session_steps == fast_before + 1  # re-established
+            res = c.batch_get(keys)
+            for i, (s_, got) in enumerate(res):
+                assert s_ == 0 and got == blobs[i], i
+            c.close()
+        finally:
+            cl.stop()
+
+    def test_host_session_not_kept_without_replace(self):
+        import numpy as np
+        cl = Cluster(n_workers=1, pool_bytes=64 << 20)
+        try:
+            c = cl.client()
+            arr = np.zeros(4096, np.uint8)
+            sess = bb.HostPutSession()
+            cfg = bb.PlacementConfig()  # replace=False
+            assert c.batch_put_session([("nk", arr)], cfg, sess) == [0]
+            assert not sess.active
+            assert c.host_session_steps == 0
+            c.close()
+        finally:
+            cl.stop()
