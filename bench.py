@@ -249,7 +249,10 @@ def main():
         cfg.replace = not no_replace
         host_sess = bb.HostPutSession()
 
+        phase_log = os.environ.get("BB_BENCH_PHASES") == "1"
+
         def do_step(lane=0):
+            tp = time.perf_counter()
             st = client.batch_put_session(put_items, cfg, host_sess)
             assert all(s == 0 for s in st), f"put failures: {st[:5]}"
             t0 = time.perf_counter()
@@ -258,6 +261,9 @@ def main():
             assert all(s == 0 for s, _ in res)
             if no_replace:
                 client.batch_remove(keys)
+            if phase_log:
+                log(f"phases put={1e3*(t0-tp):.2f} get={get_ms:.2f} "
+                    f"fast={client.host_session_steps}")
             return get_ms
         lanes = 1  # host tier: single lane
 
