@@ -249,7 +249,7 @@ void RpcClient::fail_all_pending(ErrorCode code) {
   for (auto& [id, p] : pending_) {
     p->failed = true;
     p->status = static_cast<int32_t>(code);
-    p->done = true;
+    p->done.store(true, std::memory_order_release);
   }
   cv_.notify_all();
 }
@@ -264,7 +264,7 @@ void RpcClient::reader_loop() {
       auto it = pending_.find(f.id);
       if (it != pending_.end()) {
         it->second->body = std::move(f.body);
-        it->second->done = true;
+        it->second->done.store(true, std::memory_order_release);
         cv_.notify_all();
       }
     } else if (f.kind == FrameKind::EVENT) {
@@ -313,9 +313,21 @@ Result<std::string> RpcClient::call_raw(uint16_t method, const std::string& body
     pending_.erase(id);
     return sr.error();
   }
+  // bounded spin before sleeping: the reply for a small metadata call
+  // arrives in tens of µs, and a cv sleep on an idle core costs far more
+  // to wake than the answer took to compute
+  for (int spin = 0; spin < 4000 && !p->done.load(std::memory_order_acquire);
+       ++spin) {
+#if defined(__x86_64__)
+    __builtin_ia32_pause();
+#else
+    std::this_thread::yield();
+#endif
+  }
   std::unique_lock<std::mutex> lk(mu_);
-  bool ok = cv_.wait_for(lk, std::chrono::milliseconds(timeout_ms),
-                         [&] { return p->done; });
+  bool ok = p->done.load(std::memory_order_acquire) ||
+            cv_.wait_for(lk, std::chrono::milliseconds(timeout_ms),
+                         [&] { return p->done.load(std::memory_order_acquire); });
   pending_.erase(id);
   if (!ok) return Error{ErrorCode::TIMEOUT, "rpc timeout (method " +
                                               std::to_string(method) + ")"};
