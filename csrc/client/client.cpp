@@ -550,6 +550,12 @@ Result<std::vector<int32_t>> Client::batch_remove(
 bool Client::failover_retriable(int32_t st) {
   switch (static_cast<ErrorCode>(st)) {
     case ErrorCode::OBJECT_NOT_FOUND:   // new leader never saw the PENDING put
+    case ErrorCode::INVALID_STATE:      // replace-put interrupted between
+                                        // start and commit: the new leader
+                                        // restored the PREVIOUS committed
+                                        // digest, so our commit looks like a
+                                        // different-content double-commit —
+                                        // a full redo re-places + rewrites
     case ErrorCode::NOT_LEADER:
     case ErrorCode::NOT_CONNECTED:
     case ErrorCode::CONNECT_FAILED:

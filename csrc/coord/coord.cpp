@@ -757,8 +757,12 @@ void LeaderElector::loop() {
         if (cur.ok() && cur.value() == id_) elected = true;
       }
       if (elected) {
-        leader_ = true;
+        // run the promotion hook BEFORE advertising leadership: a promoted
+        // keystone rescans persisted state in the hook, and requests that
+        // slip in between the flag flip and the rescan would see an empty
+        // object map (clients got NOT_FOUND for objects that exist)
         if (on_elected_) on_elected_();
+        leader_ = true;
       }
     } else {
       auto r = coord_->keep_alive(key_, lease_ms_);

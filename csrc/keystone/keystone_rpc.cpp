@@ -73,11 +73,15 @@ void KeystoneServer::register_handlers() {
     return serde::to_bytes(p);
   });
   rpc_.register_handler(M::OBJECT_EXISTS, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
+    if (!ks.is_leader())
+      return Error{ErrorCode::NOT_LEADER, "standby keystone; retry on leader"};
     auto r = decode<KeyMsg>(b);
     if (!r.ok()) return r.error();
     return serde::to_bytes(BoolMsg{static_cast<uint8_t>(ks.object_exists(r->key))});
   });
   rpc_.register_handler(M::GET_WORKERS, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
+    if (!ks.is_leader())
+      return Error{ErrorCode::NOT_LEADER, "standby keystone; retry on leader"};
     auto r = decode<KeyMsg>(b);
     if (!r.ok()) return r.error();
     auto resp = ks.get_workers(r->key);
@@ -85,6 +89,8 @@ void KeystoneServer::register_handlers() {
     return serde::to_bytes(resp.value());
   });
   rpc_.register_handler(M::PUT_START, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
+    if (!ks.is_leader())
+      return Error{ErrorCode::NOT_LEADER, "standby keystone; retry on leader"};
     auto r = decode<PutStartRequest>(b);
     if (!r.ok()) return r.error();
     auto resp = ks.put_start(r->key, r->size, r->config);
@@ -92,24 +98,32 @@ void KeystoneServer::register_handlers() {
     return serde::to_bytes(resp.value());
   });
   rpc_.register_handler(M::PUT_COMPLETE, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
+    if (!ks.is_leader())
+      return Error{ErrorCode::NOT_LEADER, "standby keystone; retry on leader"};
     auto r = decode<PutCompleteRequest>(b);
     if (!r.ok()) return r.error();
     BB_RETURN_IF_ERROR(ks.put_complete(r->key, r->checksum, r->shard_digests));
     return std::string{};
   });
   rpc_.register_handler(M::PUT_CANCEL, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
+    if (!ks.is_leader())
+      return Error{ErrorCode::NOT_LEADER, "standby keystone; retry on leader"};
     auto r = decode<KeyMsg>(b);
     if (!r.ok()) return r.error();
     BB_RETURN_IF_ERROR(ks.put_cancel(r->key));
     return std::string{};
   });
   rpc_.register_handler(M::REMOVE_OBJECT, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
+    if (!ks.is_leader())
+      return Error{ErrorCode::NOT_LEADER, "standby keystone; retry on leader"};
     auto r = decode<KeyMsg>(b);
     if (!r.ok()) return r.error();
     BB_RETURN_IF_ERROR(ks.remove_object(r->key));
     return std::string{};
   });
   rpc_.register_handler(M::REMOVE_ALL_OBJECTS, [&ks](const std::string&, const Ctx&) -> Result<std::string> {
+    if (!ks.is_leader())
+      return Error{ErrorCode::NOT_LEADER, "standby keystone; retry on leader"};
     return serde::to_bytes(U64Msg{ks.remove_all_objects()});
   });
   rpc_.register_handler(M::GET_WORKERS_INFO, [&ks](const std::string&, const Ctx&) -> Result<std::string> {
@@ -131,6 +145,8 @@ void KeystoneServer::register_handlers() {
     return serde::to_bytes(U64Msg{ks.get_view_version()});
   });
   rpc_.register_handler(M::LIST_OBJECTS, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
+    if (!ks.is_leader())
+      return Error{ErrorCode::NOT_LEADER, "standby keystone; retry on leader"};
     serde::Dec d(b.data(), b.size());
     std::string prefix = d.str();
     uint32_t limit = d.num<uint32_t>();
@@ -165,16 +181,22 @@ void KeystoneServer::register_handlers() {
     return serde::to_bytes(ks.batch_put_start(r->requests));
   });
   rpc_.register_handler(M::BATCH_PUT_COMPLETE, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
+    if (!ks.is_leader())
+      return Error{ErrorCode::NOT_LEADER, "standby keystone; retry on leader"};
     auto r = decode<PutCompleteListMsg>(b);
     if (!r.ok()) return r.error();
     return serde::to_bytes(StatusListMsg{ks.batch_put_complete(r->reqs)});
   });
   rpc_.register_handler(M::BATCH_PUT_CANCEL, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
+    if (!ks.is_leader())
+      return Error{ErrorCode::NOT_LEADER, "standby keystone; retry on leader"};
     auto r = decode<KeysMsg>(b);
     if (!r.ok()) return r.error();
     return serde::to_bytes(StatusListMsg{ks.batch_put_cancel(r->keys)});
   });
   rpc_.register_handler(M::BATCH_GET_WORKERS, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
+    if (!ks.is_leader())
+      return Error{ErrorCode::NOT_LEADER, "standby keystone; retry on leader"};
     auto r = decode<KeysMsg>(b);
     if (!r.ok()) return r.error();
     return serde::to_bytes(ks.batch_get_workers(r->keys));
@@ -239,6 +261,8 @@ void KeystoneServer::register_handlers() {
     return std::move(e.buf);
   });
   rpc_.register_handler(M::BATCH_GET_WORKERS2, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
+    if (!ks.is_leader())
+      return Error{ErrorCode::NOT_LEADER, "standby keystone; retry on leader"};
     serde::Dec d(b.data(), b.size());
     uint32_t count = d.num<uint32_t>();
     std::vector<std::string> keys(count);
@@ -299,11 +323,15 @@ void KeystoneServer::register_handlers() {
     return std::string{};
   });
   rpc_.register_handler(M::BATCH_REMOVE, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
+    if (!ks.is_leader())
+      return Error{ErrorCode::NOT_LEADER, "standby keystone; retry on leader"};
     auto r = decode<KeysMsg>(b);
     if (!r.ok()) return r.error();
     return serde::to_bytes(StatusListMsg{ks.batch_remove(r->keys)});
   });
   rpc_.register_handler(M::BATCH_OBJECT_EXISTS, [&ks](const std::string& b, const Ctx&) -> Result<std::string> {
+    if (!ks.is_leader())
+      return Error{ErrorCode::NOT_LEADER, "standby keystone; retry on leader"};
     auto r = decode<KeysMsg>(b);
     if (!r.ok()) return r.error();
     return serde::to_bytes(ExistsListMsg{ks.batch_object_exists(r->keys)});

@@ -1674,7 +1674,9 @@ void KeystoneService::handle_heartbeat_event(const coord::WatchEvent& ev) {
     if (it != workers_.end()) it->second.last_heartbeat_ms = now_ms();
   } else {
     // TTL expiry or explicit delete ⇒ the worker is dead
-    BB_LOG(WARN) << "worker heartbeat lost: " << id;
+    BB_LOG(WARN) << "worker heartbeat lost: " << id << " (event="
+                 << (ev.type == coord::EventType::DELETE ? "DELETE" : "EXPIRE")
+                 << " key=" << ev.key << ")";
     cleanup_dead_worker(id);
   }
 }

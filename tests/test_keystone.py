@@ -642,11 +642,32 @@ class TestFailoverMidBatch:
         kt = threading.Thread(target=killer)
         kt.start()
         try:
+            import numpy as np
             B = 32
-            blobs = [_os.urandom(32 * 1024) for _ in range(B)]
+            S = 32 * 1024
+            # half the load runs the session fast path (stable buffers,
+            # replace mode, fixed keys) — a leader kill must be absorbed by
+            # BOTH the sessionful and the plain batch paths
+            arrs = [np.zeros(S, np.uint8) for _ in range(B)]
+            sess_items = [("sess-%d" % i, arrs[i]) for i in range(B)]
+            sess_keys = [k for k, _ in sess_items]
+            scfg = bb.PlacementConfig()
+            scfg.replace = True
+            scfg.checksum = True
+            sess = bb.HostPutSession()
+            blobs = [_os.urandom(S) for _ in range(B)]
             step = 0
             end = time.time() + 4.0
             while time.time() < end or not killed.is_set():
+                sblobs = [_os.urandom(S) for _ in range(B)]
+                for a, b in zip(arrs, sblobs):
+                    a[:] = np.frombuffer(b, np.uint8)
+                st = c.batch_put_session(sess_items, scfg, sess)
+                assert all(s == 0 for s in st), ("sess", step, st[:5])
+                sres = c.batch_get(sess_keys)
+                assert all(s == 0 for s, _ in sres), ("sessget", step, [s for s, _ in sres][:6])
+                for i, (s_, v) in enumerate(sres):
+                    assert v == sblobs[i], ("sessdata", step, i)
                 keys = ["mb-%d-%d" % (step, i) for i in range(B)]
                 st = c.batch_put(list(zip(keys, blobs)))
                 assert all(s == 0 for s in st), (step, st[:5])
