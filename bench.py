@@ -325,7 +325,9 @@ def main():
     # removes). Reported as value_cold so regressions on the uncached path
     # are visible in the official record. ----
     cold_gbps = None
-    cold_steps = max(2, args.steps // 4)
+    # a regime SAMPLE, not a soak: cap it so huge --steps soaks don't spend
+    # longer in the cold phase than the timed region
+    cold_steps = max(2, min(args.steps // 4, 200))
     if not use_gpu:
         # host-tier cold regime: fresh keys, full control plane, removes
         cold_cfg = bb.PlacementConfig()
