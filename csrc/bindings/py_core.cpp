@@ -341,6 +341,17 @@ PYBIND11_MODULE(_core, m) {
       }, py::call_guard<py::gil_scoped_release>())
       .def_property_readonly("observed_epoch",
                              &coord::CoordClient::observed_epoch)
+      .def("put_many",
+           [](coord::CoordClient& c,
+              const std::vector<std::pair<std::string, std::string>>& puts,
+              const std::vector<std::string>& dels) {
+             std::vector<coord::KV> kvs;
+             kvs.reserve(puts.size());
+             for (auto& [k, v] : puts) kvs.push_back({k, v});
+             py::gil_scoped_release rel;
+             unwrap_void(c.put_many(kvs, dels));
+           },
+           py::arg("puts"), py::arg("dels") = std::vector<std::string>{})
       .def("close", &coord::CoordClient::close,
            py::call_guard<py::gil_scoped_release>());
 

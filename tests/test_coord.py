@@ -546,6 +546,9 @@ class TestFencing:
             c2.connect(ep_a2)
             with pytest.raises(Exception, match="NOT_LEADER"):
                 c2.put("/z/other", "x")
+            # the batched mutation path (durable-ack commits) is fenced too
+            with pytest.raises(Exception, match="NOT_LEADER"):
+                c2.put_many([("/z/pm", "y")], ["/z/k"])
             # the split-brain write never landed anywhere
             assert b.store().get("/z/k") == "v2"
             with pytest.raises(Exception):
