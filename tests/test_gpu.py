@@ -824,7 +824,7 @@ class TestBatchSessions:
         finally:
             cl.stop()
 
-    def test_session_not_used_when_replicated_or_cache_off(self):
+    def test_session_not_used_when_cache_off(self):
         cl = Cluster(n_workers=1, pool_bytes=256 * MB,
                      storage_class=bb.StorageClass.RAM_GPU)
         g = bb.core.gpu
