@@ -2,8 +2,6 @@
 
 #include <string.h>
 #include <sys/socket.h>
-
-#include <thread>
 #include <unistd.h>
 
 #include "blackbird/common/log.h"
@@ -32,26 +30,6 @@ Result<void> send_frame(int fd, FrameKind kind, uint64_t id, uint16_t method,
   }
   return net::write_all2(fd, &h, sizeof(h), body, body_len);
 }
-
-
-namespace {
-// Bounded busy-poll before a blocking recv: latency-bound RPC ping-pong on
-// an idle host otherwise pays the deep-idle wake latency on every message
-// (the client side spins on its response flag the same way). ~100 µs of
-// PAUSE polling per message gap, then fall back to the blocking read.
-inline void spin_for_readable(int fd) {
-  char b;
-  for (int i = 0; i < 4000; ++i) {
-    if (::recv(fd, &b, 1, MSG_PEEK | MSG_DONTWAIT) != -1 || errno != EAGAIN)
-      return;  // readable, closed, or a real error: let recv_frame see it
-#if defined(__x86_64__)
-    __builtin_ia32_pause();
-#else
-    std::this_thread::yield();
-#endif
-  }
-}
-}  // namespace
 
 Result<Frame> recv_frame(int fd) {
   WireHeader h;
@@ -157,7 +135,6 @@ void RpcServer::conn_loop(int fd, uint64_t conn_id) {
   };
 
   while (running_) {
-    spin_for_readable(fd);
     auto fr = recv_frame(fd);
     if (!fr.ok()) break;
     Frame& f = fr.value();
@@ -279,7 +256,6 @@ void RpcClient::fail_all_pending(ErrorCode code) {
 
 void RpcClient::reader_loop() {
   while (running_) {
-    spin_for_readable(fd_);
     auto fr = recv_frame(fd_);
     if (!fr.ok()) break;
     Frame& f = fr.value();
