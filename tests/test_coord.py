@@ -497,6 +497,44 @@ class TestReplication:
             b.stop()
 
 
+class TestPutManyReplication:
+    def test_put_many_streams_to_follower(self):
+        """The batched durable-ack mutation (PUT_MANY) must replicate to a
+        standby exactly like individual puts/dels — the promoted standby
+        carries every object the primary acked."""
+        a = bb.CoordServer()
+        a.start("127.0.0.1", 0)
+        ep_a = "127.0.0.1:%d" % a.port
+        b = bb.CoordServer()
+        b.start("127.0.0.1", 0)
+        c = bb.CoordClient()
+        try:
+            f = bb.CoordFollower(b, ep_a, failover_ms=60000)
+            f.start()
+            c.connect(ep_a)
+            c.put_many([("/pm/a", "1"), ("/pm/b", "2"), ("/pm/c", "3")], [])
+            c.put_many([("/pm/d", "4")], ["/pm/b"])
+            deadline = time.time() + 5
+            while time.time() < deadline:
+                try:
+                    if (b.store().get("/pm/d") == "4" and
+                            b.store().get("/pm/a") == "1"):
+                        break
+                except Exception:
+                    pass
+                time.sleep(0.05)
+            assert b.store().get("/pm/a") == "1"
+            assert b.store().get("/pm/c") == "3"
+            assert b.store().get("/pm/d") == "4"
+            with pytest.raises(Exception):
+                b.store().get("/pm/b")  # deleted in the same batch
+            f.stop()
+        finally:
+            c.close()
+            b.stop()
+            a.stop()
+
+
 class TestFencing:
     """Leadership fencing epochs: a deposed primary that
     revives AFTER standby promotion must not accept writes. Clients carry the

@@ -594,3 +594,52 @@ class TestHostSessions:
             c.close()
         finally:
             cl.stop()
+
+    def test_host_session_replicated(self):
+        """replication=2 host batches also ride the session fast path: one
+        memcpy per replica per step, shared digest on both shards, and the
+        data survives losing either worker."""
+        import numpy as np
+        cl = Cluster(n_workers=2, pool_bytes=64 << 20)
+        try:
+            c = cl.client()
+            N, S = 8, 16384
+            arrs = [np.zeros(S, np.uint8) for _ in range(N)]
+            items = [("hr%02d" % i, arrs[i]) for i in range(N)]
+            keys = [k for k, _ in items]
+            cfg = bb.PlacementConfig()
+            cfg.replace = True
+            cfg.checksum = True
+            cfg.replication = 2
+            sess = bb.HostPutSession()
+            ks = cl.keystone.service()
+            for step in range(3):
+                blobs = [os.urandom(S) for _ in range(N)]
+                for a, b in zip(arrs, blobs):
+                    a[:] = np.frombuffer(b, np.uint8)
+                assert c.batch_put_session(items, cfg, sess) == [0] * N
+            assert c.host_session_steps >= 2, c.host_session_steps
+            info = ks.get_workers(keys[2])
+            assert len(info.copies) == 2
+            assert {cp.shards[0].worker_id for cp in info.copies} == \
+                {"w0", "w1"}
+            for cp in info.copies:
+                assert cp.shards[0].digest == info.checksum
+            assert info.checksum == bb.core.gpu.checksum_cpu(blobs[2])
+            # kill either worker: the surviving replica serves the bytes
+            cl.workers[0].stop()
+            deadline = time.time() + 8
+            while time.time() < deadline:
+                try:
+                    cps = ks.get_workers(keys[2]).copies
+                    if all(cp.shards[0].worker_id != "w0" for cp in cps):
+                        break
+                except Exception:
+                    pass
+                time.sleep(0.1)
+            res = c.batch_get(keys)
+            for i, (s_, got) in enumerate(res):
+                assert s_ == 0 and got == blobs[i], i
+            c.close()
+        finally:
+            cl.stop()
