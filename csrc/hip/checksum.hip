@@ -289,6 +289,9 @@ Result<void> mfma_i8_probe(const int8_t* host_a, const int8_t* host_b,
   mfma_i8_probe_kernel<<<1, 64>>>(dA, dB, dC);
   if (hipError_t _le = hipGetLastError(); _le != hipSuccess)
     return hip_error(_le, "mfma_i8_probe_kernel launch");
+  // the result copy rides a NON-blocking stream (copy_sync) which does not
+  // order after the legacy-stream launch above — synchronize explicitly
+  BB_HIP_TRY(hipDeviceSynchronize());
   BB_RETURN_IF_ERROR(copy_sync(host_c, dC, 1024 * sizeof(int32_t), hipMemcpyDeviceToHost));
   BB_HIP_TRY(hipFree(dA));
   BB_HIP_TRY(hipFree(dB));
