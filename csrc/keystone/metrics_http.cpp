@@ -111,7 +111,13 @@ std::string MetricsHttpServer::render_metrics() {
      << "# TYPE blackbird_evictions_total counter\n"
      << "blackbird_evictions_total " << mc.evictions << "\n"
      << "# TYPE blackbird_gc_reclaimed_total counter\n"
-     << "blackbird_gc_reclaimed_total " << mc.gc_reclaimed << "\n";
+     << "blackbird_gc_reclaimed_total " << mc.gc_reclaimed << "\n"
+     << "# HELP blackbird_token_commits_total Batch commits by session "
+        "token (key-free fast-path commits)\n"
+     << "# TYPE blackbird_token_commits_total counter\n"
+     << "blackbird_token_commits_total " << ks_.token_commits() << "\n"
+     << "# TYPE blackbird_is_leader gauge\n"
+     << "blackbird_is_leader " << (ks_.is_leader() ? 1 : 0) << "\n";
   // per-pool gauges
   for (const auto& p : ks_.get_memory_pools()) {
     os << "blackbird_pool_used_bytes{pool=\"" << p.pool_id << "\",worker=\""
