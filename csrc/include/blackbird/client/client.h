@@ -71,7 +71,11 @@ class Client {
     const void* data;
     uint64_t size;
   };
-  // Host-tier batch session (the DRAM twin of GpuClient's BatchPutSession):
+  // Host-tier batch session (the DRAM twin of GpuClient's BatchPutSession).
+  // Reference analogue: the repeated batch_put_start/put_complete cycle
+  // (reference keystone_service.cpp:302-360) re-sent every key and
+  // re-allocated every step; a session pins the placements server-side and
+  // reduces the steady state to token-addressed commits.
   // callers re-putting the SAME batch (same keys/buffers/sizes, replace
   // mode) pay two tiny RPCs per step — upsert start, token+digest commit —
   // around direct memcpys into the mapped pool ranges. Established by the

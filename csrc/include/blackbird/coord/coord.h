@@ -72,7 +72,8 @@ class CoordService {
   virtual Result<uint64_t> watch_prefix(const std::string& prefix, WatchCallback cb) = 0;
   virtual Result<void> unwatch(uint64_t watch_id) = 0;
   // batched puts (ttl 0) + deletes; remote implementations send ONE round
-  // trip (synchronous durability of a whole commit batch)
+  // trip (synchronous durability of a whole commit batch — the role etcd
+  // txn batches play for the reference's metadata, etcd_service.cpp:60-86)
   virtual Result<void> put_many(const std::vector<KV>& puts,
                                 const std::vector<std::string>& dels) {
     for (const auto& kv : puts) BB_RETURN_IF_ERROR(put(kv.key, kv.value, 0));
