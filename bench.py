@@ -440,20 +440,29 @@ def main():
             shuffle_ab["ipc_gather_gbps"] = round(
                 B * S * ab_steps * WORLD / ipc_s / 1e9, 1)
 
-        # (b) RCCL all-to-all shuffle of the same batches
+        # (b) RCCL all-to-all shuffle of the same batches.
+        # ncclCommInitRank blocks until EVERY rank calls it, so the ranks
+        # agree they are ready BEFORE anyone enters it — a rank failing
+        # during setup must not leave its peers blocked inside RCCL.
         eng = coord_cl = None
         ok = True
         try:
             coord_cl = bb.CoordClient()
             coord_cl.connect(coord_ep)
             eng = bb.RcclEngine()
-            eng.init(coord_cl, "default", f"bench-ab-{base_port}", RANK,
-                     WORLD, DEVICE)
-        except Exception as e:  # e.g. ranks sharing one device (RCCL refuses)
-            log(f"shuffle A/B rccl init failed: {e}")
+        except Exception as e:
+            log(f"shuffle A/B rccl setup failed: {e}")
             shuffle_ab["rccl_error"] = str(e)[:200]
             ok = False
         if all_ranks_ok(ok):
+            try:
+                eng.init(coord_cl, "default", f"bench-ab-{base_port}", RANK,
+                         WORLD, DEVICE)
+            except Exception as e:  # e.g. ranks sharing one device
+                log(f"shuffle A/B rccl init failed: {e}")
+                shuffle_ab["rccl_error"] = str(e)[:200]
+                ok = False
+        if all_ranks_ok(ok):  # collective: every rank must participate
             wants = [([], [], 0)] * WORLD
             wants[peer] = (peer_keys, [S] * B, recv)
             bb.core.gpu_batch_shuffle(gcl, eng, wants)  # warmup
