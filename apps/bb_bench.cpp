@@ -160,9 +160,14 @@ int main(int argc, char** argv) {
     for (int b = 0; b < batch; ++b) keys.push_back("bench/b" + std::to_string(b));
     for (int b = 0; b < batch; ++b)
       items.push_back({keys[b], payload.data(), size});
+    // steady-state mode (default): in-place upserts + host batch session —
+    // two tiny RPCs per step around direct memcpys into the mapped pool
+    if (upsert && pcfg.replication <= 1) pcfg.replace = true;
+    Client::HostPutSession hsess;
     for (int it = 0; it < iters; ++it) {
       auto t0 = Clock::now();
-      auto pr = client.batch_put(items, pcfg);
+      auto pr = client.batch_put(items, pcfg,
+                                 pcfg.replace ? &hsess : nullptr);
       if (!pr.ok()) {
         std::cerr << "batch_put failed: " << pr.message() << "\n";
         return 1;
@@ -177,8 +182,9 @@ int main(int argc, char** argv) {
       }
       get_ms.push_back(ms_since(t0));
       get_bytes += static_cast<double>(size) * batch;
-      client.batch_remove(keys);
+      if (!pcfg.replace) client.batch_remove(keys);
     }
+    if (pcfg.replace) client.batch_remove(keys);
   } else {
     for (int it = 0; it < iters; ++it) {
       std::string key = "bench/k" + std::to_string(it);
