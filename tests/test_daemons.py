@@ -116,6 +116,8 @@ class TestDaemons:
         assert "blackbird_pool_capacity_bytes" in body
         assert "blackbird_repairs_total" in body
         assert "blackbird_scrub_quarantined_total" in body
+        assert "blackbird_token_commits_total" in body
+        assert "blackbird_is_leader 1" in body
         stats = json.loads(urllib.request.urlopen(
             f"http://127.0.0.1:{daemon_cluster['metrics']}/stats",
             timeout=5).read())
