@@ -643,3 +643,39 @@ class TestHostSessions:
             c.close()
         finally:
             cl.stop()
+
+    def test_host_session_survives_ttl_expiry(self):
+        """Objects under a session carry a TTL: when they expire and GC
+        reclaims them mid-session, the next step must transparently fall
+        back, re-place fresh objects and re-establish — never error."""
+        import numpy as np
+        cl = Cluster(n_workers=1, pool_bytes=64 << 20)
+        try:
+            c = cl.client()
+            N, S = 8, 8192
+            arrs = [np.zeros(S, np.uint8) for _ in range(N)]
+            items = [("tt%02d" % i, arrs[i]) for i in range(N)]
+            cfg = bb.PlacementConfig()
+            cfg.replace = True
+            cfg.checksum = True
+            cfg.ttl_ms = 300
+            sess = bb.HostPutSession()
+            ks = cl.keystone.service()
+            blobs = [os.urandom(S) for _ in range(N)]
+            for a, b in zip(arrs, blobs):
+                a[:] = np.frombuffer(b, np.uint8)
+            for step in range(3):  # establish + ride
+                assert c.batch_put_session(items, cfg, sess) == [0] * N
+            assert c.host_session_steps >= 1
+            time.sleep(0.5)  # TTL fires (each step restarted it; now lapse)
+            ks.run_gc_once()  # reclaims the expired objects
+            fast = c.host_session_steps
+            assert c.batch_put_session(items, cfg, sess) == [0] * N  # re-place
+            assert c.batch_put_session(items, cfg, sess) == [0] * N  # fast again
+            assert c.host_session_steps == fast + 1
+            res = c.batch_get([k for k, _ in items])
+            for i, (s_, got) in enumerate(res):
+                assert s_ == 0 and got == blobs[i], i
+            c.close()
+        finally:
+            cl.stop()
