@@ -679,3 +679,56 @@ class TestHostSessions:
             c.close()
         finally:
             cl.stop()
+
+    def test_host_session_survives_eviction(self):
+        """Watermark eviction may reclaim session objects (they are
+        COMMITTED between steps): the next step re-places and re-establishes
+        with zero caller-visible errors."""
+        import numpy as np
+        cs = bb.CoordServer(); cs.start("127.0.0.1", 0)
+        ep = "127.0.0.1:%d" % cs.port
+        kc = bb.KeystoneConfig()
+        kc.listen_address = "127.0.0.1:0"
+        kc.coord_endpoint = ep
+        kc.gc_interval_ms = 100000
+        kc.eviction_high_watermark = 0.5
+        kc.eviction_ratio = 0.9
+        srv = bb.create_and_start_keystone(kc)
+        wc = bb.WorkerConfig(); wc.worker_id = "evw0"; wc.coord_endpoint = ep
+        wc.data_listen_address = "127.0.0.1:0"
+        p = bb.PoolConfig(); p.pool_id = "evp"
+        p.storage_class = bb.StorageClass.RAM_CPU; p.size_bytes = 8 << 20
+        wc.pools = [p]
+        w = bb.WorkerService(wc); w.initialize(); w.start()
+        deadline = time.time() + 5
+        while time.time() < deadline and not srv.service().get_memory_pools():
+            time.sleep(0.02)
+        try:
+            o = bb.ClientOptions(); o.keystone_endpoint = srv.endpoint
+            c = bb.Client(o); c.connect()
+            N, S = 8, 64 * 1024
+            arrs = [np.zeros(S, np.uint8) for _ in range(N)]
+            items = [("ev%02d" % i, arrs[i]) for i in range(N)]
+            cfg = bb.PlacementConfig(); cfg.replace = True; cfg.checksum = True
+            sess = bb.HostPutSession()
+            blobs = [os.urandom(S) for _ in range(N)]
+            for a, b in zip(arrs, blobs):
+                a[:] = np.frombuffer(b, np.uint8)
+            for _ in range(3):
+                assert c.batch_put_session(items, cfg, sess) == [0] * N
+            # fill past the watermark with OTHER objects, then evict hard
+            filler = [("fill%d" % i, b"\xcc" * (512 * 1024)) for i in range(7)]
+            c.batch_put(filler)
+            time.sleep(0.01)
+            srv.service().run_eviction_once()
+            # session objects may be gone now: the next steps must recover
+            fast = c.host_session_steps
+            assert c.batch_put_session(items, cfg, sess) == [0] * N
+            assert c.batch_put_session(items, cfg, sess) == [0] * N
+            assert c.host_session_steps >= fast + 1
+            res = c.batch_get([k for k, _ in items])
+            for i, (s_, got) in enumerate(res):
+                assert s_ == 0 and got == blobs[i], i
+            c.close()
+        finally:
+            w.stop(); srv.stop(); srv.service().stop(); cs.stop()
