@@ -88,6 +88,9 @@ class RpcServer {
   struct Conn {
     int fd;
     std::shared_ptr<std::mutex> write_mu;
+    // guarded by write_mu: event pushes check it before sendmsg so a close
+    // can never race a write onto a (possibly reused) fd
+    std::shared_ptr<std::atomic<bool>> alive;
     std::thread thread;
   };
   std::map<uint64_t, Conn> conns_;

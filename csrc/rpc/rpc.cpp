@@ -98,6 +98,8 @@ void RpcServer::stop() {
   for (auto& [id, c] : conns) {
     ::shutdown(c.fd, SHUT_RDWR);
     if (c.thread.joinable()) c.thread.join();
+    std::lock_guard<std::mutex> g(*c.write_mu);  // no in-flight event push
+    c.alive->store(false);
     ::close(c.fd);
   }
 }
@@ -115,23 +117,33 @@ void RpcServer::accept_loop() {
     auto& c = conns_[id];
     c.fd = cfd;
     c.write_mu = std::make_shared<std::mutex>();
+    c.alive = std::make_shared<std::atomic<bool>>(true);
     c.thread = std::thread([this, cfd, id] { conn_loop(cfd, id); });
   }
 }
 
 void RpcServer::conn_loop(int fd, uint64_t conn_id) {
   std::shared_ptr<std::mutex> wmu;
+  std::shared_ptr<std::atomic<bool>> alive;
   {
     std::lock_guard<std::mutex> g(conns_mu_);
     auto it = conns_.find(conn_id);
-    if (it != conns_.end()) wmu = it->second.write_mu;
+    if (it != conns_.end()) {
+      wmu = it->second.write_mu;
+      alive = it->second.alive;
+    }
   }
   if (!wmu) return;
 
   ConnCtx ctx;
   ctx.conn_id = conn_id;
-  ctx.push = [fd, wmu](uint64_t event_id, const std::string& body) {
-    send_frame(fd, FrameKind::EVENT, event_id, 0, body.data(), body.size(), wmu.get());
+  ctx.push = [fd, wmu, alive](uint64_t event_id, const std::string& body) {
+    // hold the write mutex across the liveness check AND the send: closers
+    // take the same mutex before closing the fd
+    std::lock_guard<std::mutex> g(*wmu);
+    if (!alive->load()) return;
+    send_frame(fd, FrameKind::EVENT, event_id, 0, body.data(), body.size(),
+               nullptr);
   };
 
   while (running_) {
@@ -172,7 +184,11 @@ void RpcServer::conn_loop(int fd, uint64_t conn_id) {
   auto it = conns_.find(conn_id);
   if (it != conns_.end()) {
     it->second.thread.detach();
-    ::close(it->second.fd);
+    {
+      std::lock_guard<std::mutex> wg(*it->second.write_mu);
+      it->second.alive->store(false);
+      ::close(it->second.fd);
+    }
     conns_.erase(it);
   }
 }
