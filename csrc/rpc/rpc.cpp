@@ -248,9 +248,15 @@ void RpcClient::close() {
   join_or_detach(reader_);
   wake_dispatcher();
   join_or_detach(dispatcher_);
-  if (fd_ >= 0) {
-    ::close(fd_);
-    fd_ = -1;
+  {
+    // a user thread may still be inside call_raw's send (failover paths
+    // reconnect while other lanes issue calls): the send holds write_mu_,
+    // so closing under it can never hit a mid-sendmsg (or reused) fd
+    std::lock_guard<std::mutex> g(write_mu_);
+    if (fd_ >= 0) {
+      ::close(fd_);
+      fd_ = -1;
+    }
   }
   fail_all_pending(ErrorCode::CONNECTION_CLOSED);
 }
@@ -322,8 +328,16 @@ Result<std::string> RpcClient::call_raw(uint16_t method, const std::string& body
     std::lock_guard<std::mutex> g(mu_);
     pending_[id] = p;
   }
-  auto sr = send_frame(fd_, FrameKind::REQUEST, id, method, body.data(), body.size(),
-                       &write_mu_);
+  Result<void> sr{};
+  {
+    // fd_ read + send under ONE write_mu_ hold: serialized with close()
+    std::lock_guard<std::mutex> g(write_mu_);
+    if (fd_ < 0 || !running_)
+      sr = Error{ErrorCode::NOT_CONNECTED, "not connected"};
+    else
+      sr = send_frame(fd_, FrameKind::REQUEST, id, method, body.data(),
+                      body.size(), nullptr);
+  }
   if (!sr.ok()) {
     std::lock_guard<std::mutex> g(mu_);
     pending_.erase(id);
