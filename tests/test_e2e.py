@@ -760,3 +760,42 @@ class TestHostSessions:
             c.close()
         finally:
             cl.stop()
+
+    def test_host_session_survives_compaction(self):
+        """Pool compaction migrates committed objects to new offsets (epoch
+        bump): an established session must fall back transparently on its
+        next step and re-establish on the one after."""
+        import numpy as np
+        cl = Cluster(n_workers=1, pool_bytes=16 << 20)
+        try:
+            c = cl.client()
+            ks = cl.keystone.service()
+            N, S = 4, 256 * 1024
+            arrs = [np.zeros(S, np.uint8) for _ in range(N)]
+            items = [("cp%02d" % i, arrs[i]) for i in range(N)]
+            cfg = bb.PlacementConfig(); cfg.replace = True; cfg.checksum = True
+            sess = bb.HostPutSession()
+            # fragment the pool: interleave fillers, remove half
+            fillers = [("frag%d" % i, b"\x11" * (256 * 1024))
+                       for i in range(16)]
+            assert all(s == 0 for s in c.batch_put(fillers))
+            c.batch_remove([k for k, _ in fillers][::2])
+            blobs = [os.urandom(S) for _ in range(N)]
+            for a, b in zip(arrs, blobs):
+                a[:] = np.frombuffer(b, np.uint8)
+            for _ in range(3):
+                assert c.batch_put_session(items, cfg, sess) == [0] * N
+            assert c.host_session_steps >= 1
+            pool_id = ks.get_workers("cp00").copies[0].shards[0].pool_id
+            moved = ks.compact_pool(pool_id)  # defragment: objects relocate
+            fast = c.host_session_steps
+            assert c.batch_put_session(items, cfg, sess) == [0] * N
+            assert c.batch_put_session(items, cfg, sess) == [0] * N
+            if moved:  # placements changed: exactly one fallback step
+                assert c.host_session_steps == fast + 1
+            res = c.batch_get([k for k, _ in items])
+            for i, (s_, got) in enumerate(res):
+                assert s_ == 0 and got == blobs[i], i
+            c.close()
+        finally:
+            cl.stop()
