@@ -799,3 +799,47 @@ class TestHostSessions:
             c.close()
         finally:
             cl.stop()
+
+    def test_host_session_sustained_with_live_maintenance(self):
+        """500 session steps with the maintenance loops (GC/tiering/repair/
+        scrub cadence 200 ms) running CONCURRENTLY: every step fast-paths or
+        falls back cleanly, periodic full read-backs stay bit-exact."""
+        import numpy as np
+        cs = bb.CoordServer(); cs.start("127.0.0.1", 0)
+        ep = "127.0.0.1:%d" % cs.port
+        kc = bb.KeystoneConfig()
+        kc.listen_address = "127.0.0.1:0"
+        kc.coord_endpoint = ep
+        kc.gc_interval_ms = 200  # maintenance churns during the soak
+        srv = bb.create_and_start_keystone(kc)
+        wc = bb.WorkerConfig(); wc.worker_id = "skw0"; wc.coord_endpoint = ep
+        wc.data_listen_address = "127.0.0.1:0"
+        p = bb.PoolConfig(); p.pool_id = "skp"
+        p.storage_class = bb.StorageClass.RAM_CPU; p.size_bytes = 64 << 20
+        wc.pools = [p]
+        w = bb.WorkerService(wc); w.initialize(); w.start()
+        deadline = time.time() + 5
+        while time.time() < deadline and not srv.service().get_memory_pools():
+            time.sleep(0.02)
+        try:
+            o = bb.ClientOptions(); o.keystone_endpoint = srv.endpoint
+            c = bb.Client(o); c.connect()
+            B, S = 64, 4096
+            arrs = [np.zeros(S, np.uint8) for _ in range(B)]
+            items = [("sk%02d" % i, arrs[i]) for i in range(B)]
+            keys = [k for k, _ in items]
+            cfg = bb.PlacementConfig(); cfg.replace = True; cfg.checksum = True
+            sess = bb.HostPutSession()
+            for step in range(500):
+                if step % 50 == 0:
+                    for a in arrs:
+                        a[:] = np.frombuffer(os.urandom(S), np.uint8)
+                assert c.batch_put_session(items, cfg, sess) == [0] * B, step
+                if step % 100 == 0:
+                    res = c.batch_get(keys)
+                    for i, (s_, got) in enumerate(res):
+                        assert s_ == 0 and got == arrs[i].tobytes(), (step, i)
+            assert c.host_session_steps >= 450  # overwhelmingly fast-path
+            c.close()
+        finally:
+            w.stop(); srv.stop(); srv.service().stop(); cs.stop()
